@@ -1,0 +1,114 @@
+"""Micro-batch request aggregator.
+
+The reference serves each HTTP request synchronously on the event loop
+(reference app/main.py:42-86 — the CPU model blocks the loop; concurrency is
+pod-replica-level only). creditcore instead gathers concurrent requests into
+GPU micro-batches: a request's encoded rows are appended to a pending batch,
+which is flushed when it reaches ``max_rows`` or when the oldest request has
+waited ``max_wait_us`` — the classic latency/throughput knob (SURVEY.md §7.2
+M2).
+
+Semantics note: per-row outputs (predictions, outliers) are split back per
+request; the batch-level drift statistics are computed over the *merged*
+micro-batch, so concurrent requests in one flush share a drift block (drift
+is a batch-population statistic; merging requests gives it a larger sample).
+A lone request on an idle service gets exactly the reference semantics.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import time
+from dataclasses import dataclass, field
+
+import numpy as np
+
+
+@dataclass
+class _Pending:
+    codes: np.ndarray
+    nums: np.ndarray
+    future: asyncio.Future = field(repr=False)
+
+
+class MicroBatcher:
+    def __init__(
+        self,
+        score_arrays,  # callable(codes, nums) -> dict (engine.score_arrays)
+        max_rows: int = 8192,
+        max_wait_us: int = 300,
+    ):
+        self.score_arrays = score_arrays
+        self.max_rows = int(max_rows)
+        self.max_wait = max(0.0, max_wait_us * 1e-6)
+        self._pending: list[_Pending] = []
+        self._pending_rows = 0
+        self._event = asyncio.Event()
+        self._task: asyncio.Task | None = None
+        self._closed = False
+
+    async def start(self):
+        if self._task is None:
+            self._task = asyncio.create_task(self._loop())
+
+    async def close(self):
+        self._closed = True
+        self._event.set()
+        if self._task is not None:
+            await self._task
+            self._task = None
+
+    async def submit(self, codes: np.ndarray, nums: np.ndarray) -> dict:
+        """Submit one request's encoded rows; resolves to that request's slice
+        of the flushed batch output."""
+        fut = asyncio.get_running_loop().create_future()
+        self._pending.append(_Pending(codes, nums, fut))
+        self._pending_rows += len(codes)
+        self._event.set()
+        return await fut
+
+    async def _loop(self):
+        while not self._closed:
+            await self._event.wait()
+            self._event.clear()
+            if not self._pending:
+                continue
+            # wait window: let more requests pile in (bounded)
+            if self._pending_rows < self.max_rows and self.max_wait > 0:
+                deadline = time.perf_counter() + self.max_wait
+                while (
+                    self._pending_rows < self.max_rows
+                    and time.perf_counter() < deadline
+                ):
+                    await asyncio.sleep(0)
+            await self._flush()
+        if self._pending:
+            await self._flush()
+
+    async def _flush(self):
+        batch, self._pending = self._pending, []
+        self._pending_rows = 0
+        codes = np.concatenate([p.codes for p in batch], axis=0)
+        nums = np.concatenate([p.nums for p in batch], axis=0)
+        loop = asyncio.get_running_loop()
+        try:
+            out = await loop.run_in_executor(None, self.score_arrays, codes, nums)
+        except Exception as e:  # propagate to every waiter
+            for p in batch:
+                if not p.future.done():
+                    p.future.set_exception(e)
+            return
+        lo = 0
+        for p in batch:
+            hi = lo + len(p.codes)
+            sliced = {
+                "predictions": out["predictions"][lo:hi],
+                "outliers": out["outliers"][lo:hi],
+                "instance_score": out["instance_score"][lo:hi],
+                "p_vals": out["p_vals"],  # batch-level (shared)
+                "batch_rows": hi - lo,
+                "flush_rows": len(codes),
+            }
+            if not p.future.done():
+                p.future.set_result(sliced)
+            lo = hi

@@ -1,0 +1,164 @@
+"""CPU reference scorer over the packed flat-buffer model.
+
+This is (a) the golden reference the HIP kernels are tested against and
+(b) the no-GPU serving fallback. It deliberately uses float32 comparisons and
+arithmetic with the same semantics as csrc/kernels/score.hip, so GPU-vs-CPU
+tests can use tight tolerances; CPU-ref-vs-sklearn tests use loose tolerances
+(f32 vs f64 threshold rounding can flip a measure-zero set of branches).
+"""
+
+from __future__ import annotations
+
+import numpy as np
+
+from ..models.drift import chi2_from_counts, ks_2samp_d, ks_asymp_pvalue
+from ..pack import N_CAT, N_NUM, PackedModel
+
+
+def impute_nums(packed: PackedModel, nums: np.ndarray) -> np.ndarray:
+    nums = np.asarray(nums, dtype=np.float32)
+    out = np.where(np.isnan(nums), packed.medians[None, :], nums)
+    return np.ascontiguousarray(out, dtype=np.float32)
+
+
+def _traverse_forest(
+    nodes: np.ndarray,
+    offsets: np.ndarray,
+    value_of: "callable",
+    n_rows: int,
+) -> np.ndarray:
+    """Sum of leaf values over all trees, per row (vectorised level-by-level)."""
+    bits = nodes[:, 1].view(np.float32)
+    acc = np.zeros(n_rows, dtype=np.float64)
+    rows_all = np.arange(n_rows)
+    for t in range(len(offsets) - 1):
+        cur = np.full(n_rows, offsets[t], dtype=np.int64)
+        active = rows_all.copy()
+        while len(active):
+            nidx = cur[active]
+            feat = nodes[nidx, 0]
+            leaf = feat < 0
+            if leaf.any():
+                lr = active[leaf]
+                acc[lr] += bits[cur[lr]].astype(np.float64)
+                active = active[~leaf]
+                if not len(active):
+                    break
+                nidx = cur[active]
+                feat = nodes[nidx, 0]
+            v = value_of(active, feat)
+            thr = bits[nidx]
+            go_left = v <= thr
+            nxt = np.where(go_left, nodes[nidx, 2], nodes[nidx, 3])
+            # children indices are per-tree-relative; make absolute
+            cur[active] = offsets[t] + nxt
+    return acc
+
+
+def score_forest_cpu(
+    packed: PackedModel, codes: np.ndarray, nums_imp: np.ndarray
+) -> np.ndarray:
+    """Classifier P(default) per row = mean of leaf class-1 fractions."""
+    fc, fk = packed.feat_col, packed.feat_code
+
+    def value_of(rows: np.ndarray, feats: np.ndarray) -> np.ndarray:
+        col = fc[feats]
+        code = fk[feats]
+        cat = code >= 0
+        v = np.empty(len(rows), dtype=np.float32)
+        if cat.any():
+            v[cat] = (codes[rows[cat], col[cat]] == code[cat]).astype(np.float32)
+        if (~cat).any():
+            v[~cat] = nums_imp[rows[~cat], col[~cat]]
+        return v
+
+    s = _traverse_forest(packed.cls_nodes, packed.cls_tree_offsets, value_of, len(codes))
+    return (s / packed.cls_n_trees).astype(np.float64)
+
+
+def score_iforest_cpu(
+    packed: PackedModel, nums_imp: np.ndarray
+) -> tuple[np.ndarray, np.ndarray]:
+    """alibi instance_score (= -decision_function) and is_outlier flags."""
+
+    def value_of(rows: np.ndarray, feats: np.ndarray) -> np.ndarray:
+        return nums_imp[rows, feats]
+
+    depths = _traverse_forest(packed.if_nodes, packed.if_tree_offsets, value_of, len(nums_imp))
+    anomaly = 2.0 ** (-depths / packed.if_denom)  # Liu et al. anomaly score
+    score_samples = -anomaly  # sklearn score_samples
+    decision = score_samples - packed.if_offset  # sklearn decision_function
+    instance_score = -decision  # alibi-detect convention
+    return instance_score, (instance_score > packed.if_threshold).astype(np.float64)
+
+
+def drift_pvals_cpu(
+    packed: PackedModel, codes: np.ndarray, nums_imp: np.ndarray
+) -> np.ndarray:
+    """Per-feature p-values [N_CAT + N_NUM] in schema feature order."""
+    pvals = np.ones(N_CAT + N_NUM, dtype=np.float64)
+    for j in range(N_CAT):
+        lo, hi = packed.ref_cat_offsets[j], packed.ref_cat_offsets[j + 1]
+        nbins = hi - lo
+        bc = np.bincount(
+            np.where(codes[:, j] < 0, nbins - 1, codes[:, j]).astype(np.int64),
+            minlength=nbins,
+        )
+        pvals[j] = chi2_from_counts(packed.ref_cat_counts[lo:hi], bc)
+    for j in range(N_NUM):
+        lo, hi = packed.ref_sorted_offsets[j], packed.ref_sorted_offsets[j + 1]
+        d = ks_2samp_d(packed.ref_sorted[lo:hi].astype(np.float64), nums_imp[:, j])
+        pvals[N_CAT + j] = ks_asymp_pvalue(d, hi - lo, len(nums_imp))
+    return pvals
+
+
+def drift_stats_cpu(
+    packed: PackedModel, codes: np.ndarray, nums_imp: np.ndarray
+) -> tuple[np.ndarray, np.ndarray]:
+    """The raw statistics the GPU drift kernel produces: per-categorical batch
+    histograms (concatenated, ref_cat_offsets layout) and per-numeric K-S D."""
+    hists = np.zeros_like(packed.ref_cat_counts)
+    for j in range(N_CAT):
+        lo, hi = packed.ref_cat_offsets[j], packed.ref_cat_offsets[j + 1]
+        nbins = hi - lo
+        bc = np.bincount(
+            np.where(codes[:, j] < 0, nbins - 1, codes[:, j]).astype(np.int64),
+            minlength=nbins,
+        )
+        hists[lo:hi] = bc
+    ds = np.zeros(N_NUM, dtype=np.float32)
+    for j in range(N_NUM):
+        lo, hi = packed.ref_sorted_offsets[j], packed.ref_sorted_offsets[j + 1]
+        # float32 ref + float32 batch — same comparison dtype as the kernel
+        ds[j] = np.float32(
+            ks_2samp_d(packed.ref_sorted[lo:hi].astype(np.float64), nums_imp[:, j].astype(np.float64))
+        )
+    return hists, ds
+
+
+def pvals_from_stats(
+    packed: PackedModel, cat_hists: np.ndarray, ks_d: np.ndarray, n_batch: int
+) -> np.ndarray:
+    """Convert GPU drift statistics to p-values (shared by GPU + CPU paths)."""
+    pvals = np.ones(N_CAT + N_NUM, dtype=np.float64)
+    for j in range(N_CAT):
+        lo, hi = packed.ref_cat_offsets[j], packed.ref_cat_offsets[j + 1]
+        pvals[j] = chi2_from_counts(packed.ref_cat_counts[lo:hi], cat_hists[lo:hi])
+    for j in range(N_NUM):
+        lo, hi = packed.ref_sorted_offsets[j], packed.ref_sorted_offsets[j + 1]
+        pvals[N_CAT + j] = ks_asymp_pvalue(float(ks_d[j]), int(hi - lo), n_batch)
+    return pvals
+
+
+def score_batch_cpu(packed: PackedModel, codes: np.ndarray, nums: np.ndarray) -> dict:
+    """Full pipeline on CPU: proba + outliers + drift p-vals."""
+    nums_imp = impute_nums(packed, nums)
+    proba = score_forest_cpu(packed, codes, nums_imp)
+    iscore, outliers = score_iforest_cpu(packed, nums_imp)
+    pvals = drift_pvals_cpu(packed, codes, nums_imp)
+    return {
+        "predictions": proba,
+        "outliers": outliers,
+        "instance_score": iscore,
+        "p_vals": pvals,
+    }
