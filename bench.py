@@ -1,0 +1,201 @@
+"""creditcore flagship serving benchmark (driver contract).
+
+Measures BASELINE.json's headline metric — requests/sec for the
+credit-default /score path — on N GPUs of one node, one rank per GPU
+(data-parallel replica serving, the MI355X replacement for the reference's
+K8s replica scaling). One "request" = a batched /score call of 1024 rows
+(BASELINE config 2), scored through the full engine path: host encode →
+pinned H2D → HIP forest/iforest/drift kernels → D2H → p-value conversion.
+
+    python bench.py --gpus N --steps K --warmup W
+
+For N>1 the driver launches this under torch.distributed.run; each rank
+scores its own request stream (weak scaling). Rank 0 prints one JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REQUEST_ROWS = 1024
+BENCH_MODEL = {"n_estimators": 500, "max_depth": 16, "criterion": "gini"}
+TRAIN_ROWS = 20_000
+
+
+def _build_packed(cache_dir: str, seed: int = 2024):
+    """Train the reference-scale model once (rank 0), pack, cache as npz."""
+    from creditcore import train as T
+    from creditcore.data import make_uci_shaped_frame
+    from creditcore.models.forest import make_classifier_pipeline
+    from creditcore.pack import (
+        PackedModel,
+        pack_classifier_pipeline,
+        pack_drift,
+        pack_isolation_forest,
+    )
+    from creditcore.schema import FEATURES, TARGET
+
+    path = os.path.join(
+        cache_dir,
+        f"bench_packed_{BENCH_MODEL['n_estimators']}x{BENCH_MODEL['max_depth']}_{TRAIN_ROWS}.npz",
+    )
+    if os.path.exists(path):
+        return PackedModel.load(path), path
+    df = make_uci_shaped_frame(n_rows=TRAIN_ROWS, seed=seed)
+    pipe = make_classifier_pipeline({**BENCH_MODEL, "random_state": seed})
+    pipe.fit(df[FEATURES], df[TARGET].values.ravel())
+    drift, outlier = T.fit_detectors(df)
+    c = pack_classifier_pipeline(pipe)
+    o = pack_isolation_forest(outlier)
+    d = pack_drift(drift, c["vocabs"])
+    packed = PackedModel(**c, **o, **d)
+    packed.save(path)
+    return packed, path
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--rows", type=int, default=REQUEST_ROWS)
+    p.add_argument("--device", default="auto", choices=["auto", "cuda", "cpu"])
+    p.add_argument("--no-drift", action="store_true")
+    args = p.parse_args()
+
+    import torch
+
+    device = args.device
+    if device == "auto":
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world_size > 1
+    if distributed:
+        import torch.distributed as dist
+
+        backend = "nccl" if device == "cuda" else "gloo"
+        if device == "cuda":
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend=backend)
+
+    from creditcore.data import make_request_batch
+    from creditcore.engine import ScoringEngine
+
+    cache_dir = os.environ.get("TMPDIR", "/tmp")
+    if rank == 0:
+        packed, path = _build_packed(cache_dir)
+        print(f"[bench] rank0 packed model ready: {path}", file=sys.stderr)
+    if distributed:
+        import torch.distributed as dist
+
+        dist.barrier()
+    if rank != 0:
+        from creditcore.pack import PackedModel
+
+        _, path = None, os.path.join(
+            cache_dir,
+            f"bench_packed_{BENCH_MODEL['n_estimators']}x{BENCH_MODEL['max_depth']}_{TRAIN_ROWS}.npz",
+        )
+        packed = PackedModel.load(path)
+
+    engine = ScoringEngine(packed, device=device, device_index=local_rank)
+
+    # Pre-generate a pool of request bodies (list-of-dict JSON shapes); each
+    # timed step scores one full request through the serving compute path.
+    pool = [
+        make_request_batch(args.rows, seed=100 + rank * 16 + i)
+        for i in range(4)
+    ]
+    with_drift = not args.no_drift
+
+    def one_step(i: int):
+        out = engine.score_records(pool[i % len(pool)])
+        assert out["rows"] == args.rows
+        return out
+
+    for i in range(args.warmup):
+        one_step(i)
+
+    def sync():
+        if device == "cuda":
+            torch.cuda.synchronize()
+        if distributed:
+            import torch.distributed as dist
+
+            dist.barrier()
+        if device == "cuda":
+            torch.cuda.synchronize()
+
+    sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(i)
+    if device == "cuda":
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    sync()
+
+    # MAX elapsed over ranks → whole-job aggregate throughput
+    if distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if device == "cuda":
+            t = t.to(f"cuda:{local_rank}")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1e3
+    n_gpus = world_size if distributed else (args.gpus if device == "cuda" else 1)
+    requests_per_sec = n_gpus * args.steps / elapsed
+
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "requests/sec (whole node), credit-default /score, bs=1024",
+                    "value": round(requests_per_sec, 3),
+                    "unit": "requests/s",
+                    "n_gpus": n_gpus,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": round(ms_per_step, 4),
+                    "higher_is_better": True,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "fp32",
+                    "data": "synthetic (UCI-credit-default-shaped), random-seed-fitted model",
+                    "config": {
+                        "model": (
+                            f"RandomForest {BENCH_MODEL['n_estimators']}x"
+                            f"depth{BENCH_MODEL['max_depth']} + IForest100 + TabularDrift"
+                        ),
+                        "global_batch": n_gpus * args.rows,
+                        "request_rows": args.rows,
+                        "rows_per_sec": round(requests_per_sec * args.rows, 1),
+                        "with_drift": with_drift,
+                        "train_rows": TRAIN_ROWS,
+                        "parallelism": f"dp{n_gpus}",
+                        "device": device,
+                    },
+                }
+            )
+        )
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -86,8 +86,20 @@ class MicroBatcher:
             await self._flush()
 
     async def _flush(self):
-        batch, self._pending = self._pending, []
+        pending, self._pending = self._pending, []
         self._pending_rows = 0
+        # max_rows is a hard cap per engine call: chunk greedily by request
+        # boundaries (one oversized request still goes through whole).
+        while pending:
+            batch = [pending.pop(0)]
+            rows = len(batch[0].codes)
+            while pending and rows + len(pending[0].codes) <= self.max_rows:
+                p = pending.pop(0)
+                batch.append(p)
+                rows += len(p.codes)
+            await self._flush_one(batch)
+
+    async def _flush_one(self, batch):
         codes = np.concatenate([p.codes for p in batch], axis=0)
         nums = np.concatenate([p.nums for p in batch], axis=0)
         loop = asyncio.get_running_loop()

@@ -112,13 +112,18 @@ class ScoringEngine:
                 float(self.packed.if_threshold),
             )
             if with_drift:
+                # The K-S kernel sorts the batch column in LDS; cap the drift
+                # sample at its LDS capacity (drift is a batch-population
+                # statistic — a 16k-row sample of a larger batch is ample).
+                db = min(b, 16384)
                 cat_hist, ks_d = g["ext"].drift_stats(
-                    d_codes,
-                    d_nums,
+                    d_codes[:db],
+                    d_nums[:db],
                     g["medians"],
                     g["ref_sorted"],
                     g["rs_offsets"],
                     g["cat_offsets"],
+                    int(self.packed.ref_cat_offsets[-1]),
                 )
             proba_h = proba.to("cpu", non_blocking=True)
             iscore_h = iscore.to("cpu", non_blocking=True)
@@ -134,7 +139,7 @@ class ScoringEngine:
         }
         if with_drift:
             out["p_vals"] = cpu_ref.pvals_from_stats(
-                self.packed, cat_hist_h.numpy(), ks_d_h.numpy(), b
+                self.packed, cat_hist_h.numpy(), ks_d_h.numpy(), min(b, 16384)
             )
             out["cat_hist"] = cat_hist_h.numpy()
             out["ks_d"] = ks_d_h.numpy()

@@ -54,8 +54,11 @@ def ks_2samp_d(ref_sorted: np.ndarray, batch: np.ndarray) -> float:
     m = len(b)
     sl = np.searchsorted(ref_sorted, b, side="left") / n
     sr = np.searchsorted(ref_sorted, b, side="right") / n
-    j = np.arange(m, dtype=np.float64)
-    d = np.maximum(np.abs(sl - j / m), np.abs(sr - (j + 1) / m)).max()
+    # With ties inside the batch, F_batch's one-sided limits at value v are
+    # the run bounds (first index / last index + 1), not the per-element rank.
+    bl = np.searchsorted(b, b, side="left") / m
+    br = np.searchsorted(b, b, side="right") / m
+    d = np.maximum(np.abs(sl - bl), np.abs(sr - br)).max()
     return float(d)
 
 

@@ -1,0 +1,85 @@
+"""Training job + pyfunc checkpoint + registry tests (the reference's
+train→register task DAG and MLflow layout)."""
+
+from __future__ import annotations
+
+import os
+
+import numpy as np
+import pytest
+
+from creditcore import registry, train
+from creditcore.data import make_uci_shaped_frame
+from creditcore.schema import FEATURES
+
+
+def test_pyfunc_layout_on_disk(model_dir):
+    """Exact MLflow pyfunc layout (reference 02-register cell-12)."""
+    for rel in (
+        "MLmodel",
+        "conda.yaml",
+        "python_model.pkl",
+        "input_example.json",
+        "artifacts/classifier/model/model.pkl",
+        "artifacts/drift.pkl",
+        "artifacts/outlier.pkl",
+    ):
+        assert os.path.exists(os.path.join(model_dir, rel)), rel
+
+
+def test_mlmodel_flavor(model_dir):
+    import yaml
+
+    with open(os.path.join(model_dir, "MLmodel")) as f:
+        mlmodel = yaml.safe_load(f)
+    flavor = mlmodel["flavors"]["python_function"]
+    assert flavor["python_model"] == "python_model.pkl"
+    assert "artifacts_path" in flavor["artifacts"]
+
+
+def test_loaded_model_predicts(loaded_pyfunc, train_df):
+    out = loaded_pyfunc.predict(train_df[FEATURES].head(16))
+    assert len(out["predictions"]) == 16
+    assert set(out["feature_drift_batch"].keys()) == set(FEATURES)
+
+
+def test_register_and_resolve(model_dir, tmp_path):
+    root = str(tmp_path / "registry")
+    uri = registry.register_model(model_dir, "credit-default-uci-custom", root, tags={"k": "v"})
+    assert uri == "models:/credit-default-uci-custom/1"
+    path = registry.resolve_model_uri(uri, root)
+    assert os.path.exists(os.path.join(path, "MLmodel"))
+    # second registration bumps the version
+    uri2 = registry.register_model(model_dir, "credit-default-uci-custom", root)
+    assert uri2.endswith("/2")
+    latest = registry.resolve_model_uri("models:/credit-default-uci-custom/latest", root)
+    assert latest.endswith("2")
+
+
+def test_train_selects_best_by_roc_auc():
+    df = make_uci_shaped_frame(n_rows=1500, seed=3)
+    best = train.train_model(df=df, max_evals=3, seed=3)
+    assert "validation_roc_auc_score" in best.metrics
+    assert best.metrics["validation_roc_auc_score"] > 0.5  # better than chance
+    assert set(best.params.keys()) == {"n_estimators", "max_depth", "criterion"}
+    assert 100 <= best.params["n_estimators"] <= 999
+    assert 1 <= best.params["max_depth"] <= 24
+
+
+def test_training_handles_missing_values():
+    df = make_uci_shaped_frame(n_rows=1200, seed=5, missing_rate=0.05)
+    best = train.train_model(df=df, max_evals=1, seed=5)
+    assert np.isfinite(best.metrics["validation_accuracy_score"])
+
+
+def test_run_artifacts_written(tmp_path):
+    df = make_uci_shaped_frame(n_rows=800, seed=9)
+    runs = str(tmp_path / "runs")
+    train.train_model(df=df, max_evals=2, seed=9, runs_dir=runs)
+    run_dirs = os.listdir(runs)
+    assert len(run_dirs) == 2
+    import json
+
+    with open(os.path.join(runs, run_dirs[0], "run.json")) as f:
+        rec = json.load(f)
+    assert "params" in rec and "metrics" in rec
