@@ -362,14 +362,70 @@ def encode_batch(records_or_df, vocabs: list[list[str]]) -> tuple[np.ndarray, np
     (one-hot all-zero, matching OneHotEncoder(handle_unknown="ignore")), and
     missing -> the MISSING_CATEGORY code if it is in-vocab, else -1.
     Numeric NaNs pass through (imputed to the median inside the scorer).
-    """
-    import pandas as pd
 
-    df = (
-        records_or_df
-        if isinstance(records_or_df, pd.DataFrame)
-        else pd.DataFrame(records_or_df)
-    )
+    The list-of-dicts path (the serving hot path: parsed request bodies)
+    avoids pandas entirely — per-column C-level transposes + dict lookups.
+    """
+    if not isinstance(records_or_df, list):
+        return _encode_df(records_or_df, vocabs)
+    recs: list[dict] = records_or_df
+    ext = _native_encoder()
+    if ext is not None:
+        codes, nums = ext.encode_records(
+            recs, vocabs, CATEGORICAL_FEATURES, NUMERIC_FEATURES, MISSING_CATEGORY
+        )
+        return codes, nums
+    return _encode_records_py(recs, vocabs)
+
+
+def _native_encoder():
+    from .ops import gpu
+
+    return gpu._ext if gpu.available() else None
+
+
+def _encode_records_py(recs: list, vocabs: list[list[str]]) -> tuple[np.ndarray, np.ndarray]:
+    """Pure-Python fallback for the native encode_records (identical
+    semantics; used only when the extension isn't built)."""
+    from itertools import repeat
+    from operator import itemgetter
+
+    b = len(recs)
+    codes = np.empty((b, N_CAT), dtype=np.int16)
+    try:
+        # serving fast path: complete records (pydantic fills defaults) —
+        # C-level itemgetter transpose + map(dict.get, col, repeat(-1))
+        cat_rows = list(map(itemgetter(*CATEGORICAL_FEATURES), recs))
+        num_rows = list(map(itemgetter(*NUMERIC_FEATURES), recs))
+    except KeyError:
+        cat_rows = [tuple(r.get(c, MISSING_CATEGORY) for c in CATEGORICAL_FEATURES) for r in recs]
+        num_rows = [tuple(r.get(c) for c in NUMERIC_FEATURES) for r in recs]
+    for j, col in enumerate(zip(*cat_rows)):
+        vmap = _vocab_map(tuple(vocabs[j]))
+        codes[:, j] = np.fromiter(
+            map(vmap.get, col, repeat(np.int16(-1))), dtype=np.int16, count=b
+        )
+    nums = np.array(num_rows, dtype=np.float32).reshape(b, N_NUM)
+    return codes, nums
+
+
+_VOCAB_MAP_CACHE: dict = {}
+
+
+def _vocab_map(vocab: tuple) -> dict:
+    """code map for one categorical column; None/NaN resolve like the
+    MISSING_CATEGORY constant (SimpleImputer fill_value='missing')."""
+    m = _VOCAB_MAP_CACHE.get(vocab)
+    if m is None:
+        m = {v: np.int16(k) for k, v in enumerate(vocab)}
+        missing_code = m.get(MISSING_CATEGORY, np.int16(-1))
+        m[None] = missing_code
+        m[np.nan] = missing_code
+        _VOCAB_MAP_CACHE[vocab] = m
+    return m
+
+
+def _encode_df(df, vocabs: list[list[str]]) -> tuple[np.ndarray, np.ndarray]:
     b = len(df)
     codes = np.full((b, N_CAT), -1, dtype=np.int16)
     for j, col in enumerate(CATEGORICAL_FEATURES):

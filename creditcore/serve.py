@@ -84,7 +84,9 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         cfg: ServeConfig = state["cfg"]
         metrics: Metrics = state["metrics"]
         request_id = uuid.uuid4().hex
-        records = [r.model_dump() for r in data]
+        # pydantic v2 keeps validated fields in __dict__; avoids the
+        # per-request model_dump() copy on the hot path
+        records = [r.__dict__ for r in data]
 
         if cfg.log_inference_data:
             reqlog.log_inference_data(
