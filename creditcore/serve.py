@@ -23,6 +23,7 @@ import uuid
 from contextlib import asynccontextmanager
 from itertools import count
 
+import numpy as np
 from fastapi import FastAPI, HTTPException
 
 from .config import ServeConfig
@@ -50,13 +51,64 @@ def _build_engines(cfg: ServeConfig) -> list[ScoringEngine]:
     return engines
 
 
+class ReplicaPool:
+    """Per-GPU replica health + dispatch (the in-node analog of K8s dropping
+    a crashed pod from the Service — SURVEY.md §5.3). A replica is removed
+    from rotation after ``max_failures`` consecutive scoring errors; scoring
+    keeps flowing to the survivors."""
+
+    def __init__(self, n: int, max_failures: int = 3):
+        self.alive = [True] * n
+        self.fails = [0] * n
+        self._rr = count()
+
+    def pick(self) -> int:
+        n = len(self.alive)
+        for _ in range(n):
+            i = next(self._rr) % n
+            if self.alive[i]:
+                return i
+        raise RuntimeError("no healthy replicas")
+
+    def report_ok(self, i: int) -> None:
+        self.fails[i] = 0
+
+    def report_fail(self, i: int, max_failures: int = 3) -> None:
+        self.fails[i] += 1
+        if self.fails[i] >= max_failures:
+            self.alive[i] = False
+
+
 @asynccontextmanager
 async def lifespan(app: FastAPI):
+    import torch
+
+    from .parallel import DriftSync
+
     cfg: ServeConfig = app.state.cfg
     engines = _build_engines(cfg)
+    # One shared node-level drift accumulator (host-side; 23×bins int64 —
+    # the per-request drift stats feeding it come from the HIP kernels).
+    drift_sync = DriftSync(engines[0].packed, device="cpu")
+
+    def scorer(e: ScoringEngine):
+        def run(codes, nums):
+            out = e.score_arrays(codes, nums)
+            if "cat_hist" in out:
+                drift_sync.accumulate(
+                    torch.from_numpy(np.ascontiguousarray(out["cat_hist"])),
+                    torch.from_numpy(nums),
+                )
+                if drift_sync.batches % max(cfg.drift_sync_period, 1) == 0:
+                    drift_sync.allreduce()
+                    state["metrics"].observe_drift_sync()
+            return out
+
+        return run
+
     batchers = [
         MicroBatcher(
-            e.score_arrays, max_rows=cfg.max_batch_rows, max_wait_us=cfg.batch_wait_us
+            scorer(e), max_rows=cfg.max_batch_rows, max_wait_us=cfg.batch_wait_us
         )
         for e in engines
     ]
@@ -64,7 +116,8 @@ async def lifespan(app: FastAPI):
         await b.start()
     state["engines"] = engines
     state["batchers"] = batchers
-    state["rr"] = count()
+    state["pool"] = ReplicaPool(len(engines))
+    state["drift_sync"] = drift_sync
     state["metrics"] = Metrics()
     state["cfg"] = cfg
     yield
@@ -95,18 +148,26 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
 
         engines = state["engines"]
         batchers = state["batchers"]
-        idx = next(state["rr"]) % len(engines)
-        codes, nums = encode_batch(records, engines[idx].packed.vocabs)
+        pool: ReplicaPool = state["pool"]
+        try:
+            idx = pool.pick()
+        except RuntimeError:
+            metrics.observe_error()
+            raise HTTPException(status_code=503, detail="no healthy replicas")
+        try:
+            codes, nums = encode_batch(records, engines[idx].packed.vocabs)
+        except (ValueError, TypeError) as e:
+            raise HTTPException(status_code=422, detail=f"bad record: {e}")
 
         t0 = time.perf_counter()
         try:
             out = await batchers[idx].submit(codes, nums)
+            pool.report_ok(idx)
         except Exception as e:
             metrics.observe_error()
+            pool.report_fail(idx)
             raise HTTPException(status_code=500, detail=f"scoring failed: {e}")
         latency_ms = (time.perf_counter() - t0) * 1e3
-
-        import numpy as np
 
         response = {
             "predictions": [float(x) for x in out["predictions"]],
@@ -139,16 +200,27 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
 
     @app.get("/healthz")
     async def healthz():
+        pool: ReplicaPool = state["pool"]
+        any_alive = any(pool.alive)
         return {
-            "status": "ok",
+            "status": "ok" if any_alive else "dead",
             "engines": [
-                {"device": e.device, "index": e.device_index} for e in state["engines"]
+                {"device": e.device, "index": e.device_index, "alive": pool.alive[i]}
+                for i, e in enumerate(state["engines"])
             ],
         }
 
     @app.get("/metrics")
     async def metrics_endpoint():
         return state["metrics"].snapshot()
+
+    @app.get("/drift")
+    async def drift_endpoint():
+        """Node-global drift state (the cross-replica monitor the reference's
+        per-pod-isolated drift lacks — SURVEY.md §2.4)."""
+        ds = state["drift_sync"]
+        ds.allreduce()
+        return ds.snapshot()
 
     return app
 

@@ -1,0 +1,86 @@
+"""Replica health / failure-detection tests and the
+node-global drift monitor endpoint."""
+
+from __future__ import annotations
+
+import pytest
+from fastapi.testclient import TestClient
+
+from creditcore.config import ServeConfig
+from creditcore.schema import SAMPLE_REQUEST
+from creditcore.serve import ReplicaPool, create_app, state
+
+
+def test_pool_round_robin():
+    pool = ReplicaPool(3)
+    picks = [pool.pick() for _ in range(6)]
+    assert picks == [0, 1, 2, 0, 1, 2]
+
+
+def test_pool_drops_failed_replica():
+    pool = ReplicaPool(2)
+    for _ in range(3):
+        pool.report_fail(0)
+    assert pool.alive == [False, True]
+    assert {pool.pick() for _ in range(4)} == {1}
+
+
+def test_pool_ok_resets_fail_count():
+    pool = ReplicaPool(1)
+    pool.report_fail(0)
+    pool.report_fail(0)
+    pool.report_ok(0)
+    pool.report_fail(0)
+    assert pool.alive == [True]
+
+
+def test_pool_all_dead_raises():
+    pool = ReplicaPool(1)
+    for _ in range(3):
+        pool.report_fail(0)
+    with pytest.raises(RuntimeError):
+        pool.pick()
+
+
+@pytest.fixture()
+def client(model_dir):
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    cfg.drift_sync_period = 1
+    app = create_app(cfg)
+    with TestClient(app) as c:
+        yield c
+
+
+def test_drift_endpoint(client):
+    from creditcore.data import make_request_batch
+
+    client.post("/predict", json=make_request_batch(128, seed=2))
+    snap = client.get("/drift").json()
+    assert snap["rows"] >= 128
+    assert len(snap["node_feature_drift"]) == 23
+
+
+def test_dead_replicas_give_503(client):
+    pool = state["pool"]
+    for _ in range(3):
+        pool.report_fail(0)
+    r = client.post("/predict", json=SAMPLE_REQUEST)
+    assert r.status_code == 503
+    assert client.get("/healthz").json()["status"] == "dead"
+
+
+def test_engine_failure_marks_replica(client, monkeypatch):
+    """Three consecutive engine faults drop the replica from rotation."""
+    batcher = state["batchers"][0]
+
+    def boom(codes, nums):
+        raise RuntimeError("injected HIP fault")
+
+    monkeypatch.setattr(batcher, "score_arrays", boom)
+    for _ in range(3):
+        r = client.post("/predict", json=SAMPLE_REQUEST)
+        assert r.status_code == 500
+    assert state["pool"].alive == [False]
+    assert client.post("/predict", json=SAMPLE_REQUEST).status_code == 503
