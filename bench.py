@@ -91,21 +91,17 @@ def main():
     from creditcore.engine import ScoringEngine
 
     cache_dir = os.environ.get("TMPDIR", "/tmp")
+    packed = None
     if rank == 0:
         packed, path = _build_packed(cache_dir)
         print(f"[bench] rank0 packed model ready: {path}", file=sys.stderr)
     if distributed:
-        import torch.distributed as dist
+        # RCCL weight broadcast over xGMI: rank 0 distributes the packed
+        # buffers to every replica (creditcore.parallel).
+        from creditcore.parallel import broadcast_packed
 
-        dist.barrier()
-    if rank != 0:
-        from creditcore.pack import PackedModel
-
-        _, path = None, os.path.join(
-            cache_dir,
-            f"bench_packed_{BENCH_MODEL['n_estimators']}x{BENCH_MODEL['max_depth']}_{TRAIN_ROWS}.npz",
-        )
-        packed = PackedModel.load(path)
+        bdev = f"cuda:{local_rank}" if device == "cuda" else "cpu"
+        packed = broadcast_packed(packed, device=bdev, src=0)
 
     engine = ScoringEngine(packed, device=device, device_index=local_rank)
 
