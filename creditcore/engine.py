@@ -168,16 +168,16 @@ class ScoringEngine:
         t0 = time.perf_counter()
         raw = self.score_arrays(codes, nums)
         latency_ms = (time.perf_counter() - t0) * 1e3
+        # (1 - p_val) computed in float32, like the reference's
+        # (1 - drift_results["data"]["p_val"]).tolist() on the float32
+        # p_val array (02-register cell-9); .tolist() is C-speed.
+        one_minus = (
+            np.float32(1.0) - np.asarray(raw["p_vals"], dtype=np.float32)
+        ).astype(np.float64)
         resp = {
-            "predictions": [float(x) for x in raw["predictions"]],
-            "outliers": [float(x) for x in raw["outliers"]],
-            "feature_drift_batch": {
-                # (1 - p_val) in float32, like the reference's
-                # (1 - drift_results["data"]["p_val"]).tolist() on the
-                # float32 p_val array (02-register cell-9)
-                f: float(np.float32(1.0) - np.float32(p))
-                for f, p in zip(FEATURES, raw["p_vals"])
-            },
+            "predictions": np.asarray(raw["predictions"]).tolist(),
+            "outliers": np.asarray(raw["outliers"]).tolist(),
+            "feature_drift_batch": dict(zip(FEATURES, one_minus.tolist())),
         }
         return {"response": resp, "latency_ms": latency_ms, "rows": len(codes)}
 

@@ -62,6 +62,130 @@ def ks_2samp_d(ref_sorted: np.ndarray, batch: np.ndarray) -> float:
     return float(d)
 
 
+def ks_asymp_pvalue_many(ds: np.ndarray, n_ref: int, n_batch: int) -> np.ndarray:
+    """Vectorized ks_asymp_pvalue for equal sample sizes across features.
+
+    scipy's kstwo.sf costs ~0.1 ms *per value* (exact Pomeranz/Durbin path) —
+    ~1.5 ms for the 14 numeric features, which would dominate request
+    latency. For en >= 300 the Pelz-Good series agrees with the exact value
+    to < 3e-7 absolute (verified in tests at en = 974: < 1e-8), so the hot
+    path uses a vectorized Pelz-Good; smaller en falls back to exact scipy.
+    """
+    en = int(np.round(float(n_ref) * float(n_batch) / (float(n_ref) + float(n_batch))))
+    ds = np.asarray(ds, dtype=np.float64)
+    if en >= 300:
+        return _pelz_good_sf(ds, en)
+    return np.clip(stats.kstwo.sf(ds, en), 0.0, 1.0)
+
+
+_PI2 = np.pi**2
+_PI4 = np.pi**4
+_PI6 = np.pi**6
+_SQRT2PI = np.sqrt(2 * np.pi)
+_SQRT3 = np.sqrt(3.0)
+
+
+def _pelz_good_sf(xs: np.ndarray, n: int) -> np.ndarray:
+    """Vectorized Pelz-Good approximation of the one-sample K-S survival
+    function P(D_n > x) (Pelz & Good 1976; series form as in Simard &
+    L'Ecuyer 2011): Prob(Dn <= x) ~ K0(z) + K1(z)/sqrt(n) + K2(z)/n +
+    K3(z)/n^1.5 with z = x*sqrt(n), each K_i expressed through Jacobi theta
+    series that converge fast for the z values drift testing produces."""
+    xs = np.atleast_1d(xs).astype(np.float64)
+    out = np.empty_like(xs)
+    out[xs <= 0.0] = 1.0
+    out[xs >= 1.0] = 0.0
+    sel = (xs > 0.0) & (xs < 1.0)
+    if not sel.any():
+        return np.clip(out, 0.0, 1.0)
+    z = np.sqrt(n) * xs[sel]
+    z2, z3, z4, z6, z7, z8, z10 = z**2, z**3, z**4, z**6, z**7, z**8, z**10
+
+    qlog = -_PI2 / 8.0 / z2
+    tiny = qlog < -690.0  # exp underflow => cdf ~ 0 => sf ~ 1
+    q = np.exp(np.where(tiny, -690.0, qlog))
+
+    k1a = -z2
+    k1b = _PI2 / 4.0
+    k2a = 6 * z6 + 2 * z4
+    k2b = (2 * z4 - 5 * z2) * _PI2 / 4.0
+    k2c = _PI4 * (1 - 2 * z2) / 16.0
+    k3d = _PI6 * (5 - 30 * z2) / 64.0
+    k3c = _PI4 * (-60 * z2 + 212 * z4) / 16.0
+    k3b = _PI2 * (135 * z4 - 96 * z6) / 4.0
+    k3a = -30 * z6 - 90 * z8
+
+    K0 = np.zeros_like(z)
+    K1 = np.zeros_like(z)
+    K2 = np.zeros_like(z)
+    K3 = np.zeros_like(z)
+    maxk = int(np.ceil(16 * z.max() / np.pi))
+    for k in range(maxk, 0, -1):  # Horner over odd m = 2k-1 in powers of q
+        m = 2 * k - 1
+        m2, m4, m6 = m**2, m**4, m**6
+        qp = q ** (8 * k)
+        K0 = K0 * qp + 1.0
+        K1 = K1 * qp + (k1a + k1b * m2)
+        K2 = K2 * qp + (k2a + k2b * m2 + k2c * m4)
+        K3 = K3 * qp + (k3a + k3b * m2 + k3c * m4 + k3d * m6)
+    K0 *= q * _SQRT2PI / z
+    K1 *= q * _SQRT2PI / (6 * z4)
+    K2 *= q * _SQRT2PI / (72 * z7)
+    K3 *= q * _SQRT2PI / (6480 * z10)
+
+    # extra integer-k theta sums for K2, K3
+    q2 = np.exp(-_PI2 / 2.0 / z2)
+    ks = np.arange(1, maxk + 1, dtype=np.float64)
+    k2_ = ks**2
+    qpw = q2[:, None] ** k2_[None, :]
+    K2 += (qpw @ k2_) * _PI2 * _SQRT2PI / (-36 * z3)
+    kspi = np.pi * ks
+    sqrt3z = _SQRT3 * z
+    term = (sqrt3z[:, None] + kspi[None, :]) * (sqrt3z[:, None] - kspi[None, :])
+    K3 += ((term * qpw) @ k2_) * _PI2 * _SQRT2PI / (216 * z6)
+
+    cdf = K0 + K1 / np.sqrt(n) + K2 / n + K3 / n**1.5
+    cdf = np.where(tiny, 0.0, cdf)
+    out[sel] = 1.0 - cdf
+    return np.clip(out, 0.0, 1.0)
+
+
+def chi2_from_counts_many(
+    ref_counts: np.ndarray, batch_counts: np.ndarray, offsets: np.ndarray
+) -> np.ndarray:
+    """Vectorized chi2_from_counts over concatenated per-feature count
+    buffers (``offsets`` delimits features). One scipy chi2.sf call total;
+    numerically identical to scipy.stats.chi2_contingency per feature
+    (including the Yates continuity correction on 2x2 tables)."""
+    nf = len(offsets) - 1
+    stat = np.zeros(nf)
+    dof = np.zeros(nf, dtype=np.int64)
+    for j in range(nf):
+        rc = np.asarray(ref_counts[offsets[j] : offsets[j + 1]], dtype=np.float64)
+        bc = np.asarray(batch_counts[offsets[j] : offsets[j + 1]], dtype=np.float64)
+        keep = (rc + bc) > 0
+        rc, bc = rc[keep], bc[keep]
+        k = len(rc)
+        if k < 2 or rc.sum() == 0 or bc.sum() == 0:
+            dof[j] = 0
+            continue
+        n = rc.sum() + bc.sum()
+        exp_r = (rc + bc) * (rc.sum() / n)
+        exp_b = (rc + bc) * (bc.sum() / n)
+        dr = np.abs(rc - exp_r)
+        db = np.abs(bc - exp_b)
+        if k == 2:  # Yates continuity correction, as chi2_contingency applies
+            dr = np.maximum(dr - 0.5, 0.0)
+            db = np.maximum(db - 0.5, 0.0)
+        stat[j] = (dr**2 / exp_r).sum() + (db**2 / exp_b).sum()
+        dof[j] = k - 1
+    pv = np.ones(nf)
+    live = dof > 0
+    if live.any():
+        pv[live] = stats.chi2.sf(stat[live], dof[live])
+    return pv
+
+
 def chi2_from_counts(ref_counts: np.ndarray, batch_counts: np.ndarray) -> float:
     """Chi-square p-value from a 2 x k contingency table of counts.
 

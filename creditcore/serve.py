@@ -169,13 +169,13 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
             raise HTTPException(status_code=500, detail=f"scoring failed: {e}")
         latency_ms = (time.perf_counter() - t0) * 1e3
 
+        one_minus = (
+            np.float32(1.0) - np.asarray(out["p_vals"], dtype=np.float32)
+        ).astype(np.float64)
         response = {
-            "predictions": [float(x) for x in out["predictions"]],
-            "outliers": [float(x) for x in out["outliers"]],
-            "feature_drift_batch": {
-                f: float(np.float32(1.0) - np.float32(p))
-                for f, p in zip(FEATURES, out["p_vals"])
-            },
+            "predictions": np.asarray(out["predictions"]).tolist(),
+            "outliers": np.asarray(out["outliers"]).tolist(),
+            "feature_drift_batch": dict(zip(FEATURES, one_minus.tolist())),
         }
         metrics.observe_request(len(records), latency_ms)
         reqlog.log_model_output(

@@ -44,6 +44,38 @@ def test_ks_with_ties():
     assert abs(d - sp.statistic) < 1e-12
 
 
+def test_pelz_good_matches_exact_kstwo():
+    """The serving-path vectorized Pelz-Good p-value vs scipy's exact
+    kstwo.sf across the drift-relevant range."""
+    from creditcore.models.drift import _pelz_good_sf, ks_asymp_pvalue_many
+
+    for en in (300, 974, 5000):
+        ds = np.linspace(0.002, 0.9, 150)
+        np.testing.assert_allclose(
+            _pelz_good_sf(ds, en), stats.kstwo.sf(ds, en), atol=3e-7
+        )
+    # below the accuracy cutoff it must be scipy-exact
+    ds = np.array([0.05, 0.2, 0.4])
+    got = ks_asymp_pvalue_many(ds, 1000, 100)  # en = 91 < 300
+    en = round(1000 * 100 / 1100)
+    np.testing.assert_allclose(got, stats.kstwo.sf(ds, en), atol=1e-14)
+
+
+def test_vectorized_pvals_match_scalar(packed):
+    """pvals_from_stats (vectorized serving path) vs the scalar exact path."""
+    from creditcore.data import make_request_batch
+    from creditcore.ops import cpu_ref
+    from creditcore.pack import encode_batch
+
+    recs = make_request_batch(512, seed=42)
+    codes, nums = encode_batch(recs, packed.vocabs)
+    nums_imp = cpu_ref.impute_nums(packed, nums)
+    hist, ks_d = cpu_ref.drift_stats_cpu(packed, codes, nums_imp)
+    fast = cpu_ref.pvals_from_stats(packed, hist, ks_d, 512)
+    exact = cpu_ref.drift_pvals_cpu(packed, codes, nums_imp)
+    np.testing.assert_allclose(fast, exact, atol=1e-6)
+
+
 def test_chi2_matches_scipy():
     rc = np.array([50, 30, 20, 5])
     bc = np.array([10, 25, 3, 1])
