@@ -44,105 +44,81 @@ class ScoringEngine:
             self._init_gpu()
 
     # ------------------------------------------------------------------ GPU
-    def _init_gpu(self):
+    DRIFT_MAX_ROWS = 16384  # K-S kernel LDS sort capacity (csrc MAX_DRIFT_ROWS)
+
+    def _init_gpu(self, capacity: int = 16384):
         import torch
 
         from .ops import gpu
 
         ext = gpu.ext()  # raises loudly if the HIP extension is missing
-        dev = torch.device("cuda", self.device_index)
         p = self.packed
-
-        def up(a, dtype):
-            return torch.from_numpy(np.ascontiguousarray(a)).to(dtype).to(dev)
-
-        g = {
-            "torch": torch,
-            "ext": ext,
-            "dev": dev,
-            "stream": torch.cuda.Stream(device=dev),
-            "cls_nodes": up(p.cls_nodes, torch.int32),
-            "cls_offsets": up(p.cls_tree_offsets, torch.int32),
-            "if_nodes": up(p.if_nodes, torch.int32),
-            "if_offsets": up(p.if_tree_offsets, torch.int32),
-            "feat_col": up(p.feat_col, torch.int32),
-            "feat_code": up(p.feat_code, torch.int32),
-            "medians": up(p.medians, torch.float32),
-            "ref_sorted": up(p.ref_sorted, torch.float32),
-            "rs_offsets": up(p.ref_sorted_offsets, torch.int32),
-            "cat_offsets": up(p.ref_cat_offsets, torch.int32),
+        model = {
+            "cls_nodes": torch.from_numpy(np.ascontiguousarray(p.cls_nodes)),
+            "cls_tree_offsets": torch.from_numpy(p.cls_tree_offsets),
+            "feat_col": torch.from_numpy(p.feat_col),
+            "feat_code": torch.from_numpy(p.feat_code),
+            "medians": torch.from_numpy(p.medians),
+            "if_nodes": torch.from_numpy(np.ascontiguousarray(p.if_nodes)),
+            "if_tree_offsets": torch.from_numpy(p.if_tree_offsets),
+            "ref_sorted": torch.from_numpy(p.ref_sorted),
+            "ref_sorted_offsets": torch.from_numpy(p.ref_sorted_offsets),
+            "ref_cat_offsets": torch.from_numpy(p.ref_cat_offsets),
+            "if_denom": float(p.if_denom),
+            "if_offset": float(p.if_offset),
+            "if_threshold": float(p.if_threshold),
         }
-        # pinned staging (grown on demand)
-        g["pin_codes"] = None
-        g["pin_nums"] = None
-        self._gpu = g
+        sess = ext.ScoreSession(model, capacity, self.device_index)
+        self._gpu = {
+            "ext": ext,
+            "model": model,
+            "sess": sess,
+            # zero-copy numpy views of the session's pinned staging/outputs
+            "np_codes": sess.pin_codes.numpy(),
+            "np_nums": sess.pin_nums.numpy(),
+            "np_outs": sess.pin_outs.numpy(),
+            "np_hist": sess.pin_hist.numpy(),
+            "np_ksd": sess.pin_ksd.numpy(),
+        }
 
-    def _ensure_staging(self, b: int):
-        import torch
-
+    def _ensure_capacity(self, b: int):
         g = self._gpu
-        if g["pin_codes"] is None or g["pin_codes"].shape[0] < b:
-            cap = max(1024, 1 << (b - 1).bit_length())
-            g["pin_codes"] = torch.empty((cap, N_CAT), dtype=torch.int16, pin_memory=True)
-            g["pin_nums"] = torch.empty((cap, N_NUM), dtype=torch.float32, pin_memory=True)
+        if b > g["sess"].capacity:
+            cap = 1 << (b - 1).bit_length()
+            sess = g["ext"].ScoreSession(g["model"], cap, self.device_index)
+            g["sess"] = sess
+            g["np_codes"] = sess.pin_codes.numpy()
+            g["np_nums"] = sess.pin_nums.numpy()
+            g["np_outs"] = sess.pin_outs.numpy()
+            g["np_hist"] = sess.pin_hist.numpy()
+            g["np_ksd"] = sess.pin_ksd.numpy()
 
     def _score_gpu(self, codes: np.ndarray, nums: np.ndarray, with_drift: bool = True) -> dict:
-        torch = self._gpu["torch"]
         g = self._gpu
         b = len(codes)
-        self._ensure_staging(b)
-        g["pin_codes"][:b].copy_(torch.from_numpy(codes))
-        g["pin_nums"][:b].copy_(torch.from_numpy(nums))
-        with torch.cuda.stream(g["stream"]):
-            d_codes = g["pin_codes"][:b].to(g["dev"], non_blocking=True)
-            d_nums = g["pin_nums"][:b].to(g["dev"], non_blocking=True)
-            proba, iscore, outlier = g["ext"].score_forest_pipeline(
-                d_codes,
-                d_nums,
-                g["cls_nodes"],
-                g["cls_offsets"],
-                g["feat_col"],
-                g["feat_code"],
-                g["medians"],
-                int(self.packed.n_onehot),
-                g["if_nodes"],
-                g["if_offsets"],
-                float(self.packed.if_denom),
-                float(self.packed.if_offset),
-                float(self.packed.if_threshold),
-            )
-            if with_drift:
-                # The K-S kernel sorts the batch column in LDS; cap the drift
-                # sample at its LDS capacity (drift is a batch-population
-                # statistic — a 16k-row sample of a larger batch is ample).
-                db = min(b, 16384)
-                cat_hist, ks_d = g["ext"].drift_stats(
-                    d_codes[:db],
-                    d_nums[:db],
-                    g["medians"],
-                    g["ref_sorted"],
-                    g["rs_offsets"],
-                    g["cat_offsets"],
-                    int(self.packed.ref_cat_offsets[-1]),
-                )
-            proba_h = proba.to("cpu", non_blocking=True)
-            iscore_h = iscore.to("cpu", non_blocking=True)
-            outlier_h = outlier.to("cpu", non_blocking=True)
-            if with_drift:
-                cat_hist_h = cat_hist.to("cpu", non_blocking=True)
-                ks_d_h = ks_d.to("cpu", non_blocking=True)
-        g["stream"].synchronize()
+        self._ensure_capacity(b)
+        # Drift is a batch-population statistic; cap its sample at the K-S
+        # kernel's LDS sort capacity (predictions still cover every row).
+        drift_now = with_drift and b <= self.DRIFT_MAX_ROWS
+        g["np_codes"][:b] = codes
+        g["np_nums"][:b] = nums
+        g["sess"].score(b, drift_now, True)  # blocks; GIL released
         out = {
-            "predictions": proba_h.double().numpy(),
-            "outliers": outlier_h.double().numpy(),
-            "instance_score": iscore_h.double().numpy(),
+            "predictions": g["np_outs"][0, :b].copy(),
+            "instance_score": g["np_outs"][1, :b].copy(),
+            "outliers": g["np_outs"][2, :b].copy(),
         }
-        if with_drift:
-            out["p_vals"] = cpu_ref.pvals_from_stats(
-                self.packed, cat_hist_h.numpy(), ks_d_h.numpy(), min(b, 16384)
-            )
-            out["cat_hist"] = cat_hist_h.numpy()
-            out["ks_d"] = ks_d_h.numpy()
+        if with_drift and not drift_now:
+            # oversized batch: score a capped drift sample in a second pass
+            db = self.DRIFT_MAX_ROWS
+            g["sess"].score(db, True, True)
+            drift_now, b = True, db
+        if drift_now:
+            hist = g["np_hist"].copy()
+            ks_d = g["np_ksd"].copy()
+            out["p_vals"] = cpu_ref.pvals_from_stats(self.packed, hist, ks_d, b)
+            out["cat_hist"] = hist
+            out["ks_d"] = ks_d
         return out
 
     # ------------------------------------------------------------------ API

@@ -28,6 +28,7 @@
 
 #include <torch/extension.h>
 
+#include <c10/hip/HIPGuard.h>
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
@@ -376,6 +377,150 @@ std::vector<torch::Tensor> drift_stats(
 }
 
 // ---------------------------------------------------------------------------
+// ScoreSession — one replica's resident GPU state + a single-call scoring
+// path. Owns the model buffers in HBM, a private HIP stream, device
+// workspace and pinned staging, so one Python call per request performs:
+//   pinned H2D -> zero accumulators -> forest x2 -> finalize -> drift -> D2H
+// with no per-call torch dispatch or allocation (the Python-side hot loop
+// was ~10x the kernel time). The GIL is released while waiting.
+// ---------------------------------------------------------------------------
+
+struct ScoreSession {
+  torch::Tensor cls_nodes, cls_off, feat_col, feat_code, medians;
+  torch::Tensor if_nodes, if_off, ref_sorted, rs_off, cat_off;
+  torch::Tensor d_codes, d_nums, acc, outs, hist, ksd;
+  torch::Tensor pin_codes, pin_nums, pin_outs, pin_hist, pin_ksd;
+  double if_denom{}, if_offset{}, if_threshold{};
+  int64_t total_bins{}, t_cls{}, t_if{}, capacity{};
+  int device_index{};
+  hipStream_t stream{};
+
+  ScoreSession(py::dict model, int64_t cap, int dev) : capacity(cap), device_index(dev) {
+    c10::hip::HIPGuard guard((c10::DeviceIndex)dev);
+    auto devopt = torch::TensorOptions().device(torch::kCUDA, dev);
+    auto up_i32 = [&](const char* k) {
+      return py::cast<torch::Tensor>(model[k]).to(devopt.dtype(torch::kInt32)).contiguous();
+    };
+    auto up_f32 = [&](const char* k) {
+      return py::cast<torch::Tensor>(model[k]).to(devopt.dtype(torch::kFloat32)).contiguous();
+    };
+    cls_nodes = up_i32("cls_nodes");
+    cls_off = up_i32("cls_tree_offsets");
+    feat_col = up_i32("feat_col");
+    feat_code = up_i32("feat_code");
+    medians = up_f32("medians");
+    if_nodes = up_i32("if_nodes");
+    if_off = up_i32("if_tree_offsets");
+    ref_sorted = up_f32("ref_sorted");
+    rs_off = up_i32("ref_sorted_offsets");
+    cat_off = up_i32("ref_cat_offsets");
+    if_denom = py::cast<double>(model["if_denom"]);
+    if_offset = py::cast<double>(model["if_offset"]);
+    if_threshold = py::cast<double>(model["if_threshold"]);
+    t_cls = cls_off.size(0) - 1;
+    t_if = if_off.size(0) - 1;
+    total_bins = py::cast<torch::Tensor>(model["ref_cat_offsets"])[-1].item<int64_t>();
+
+    d_codes = torch::empty({capacity, N_CAT}, devopt.dtype(torch::kInt16));
+    d_nums = torch::empty({capacity, N_NUM}, devopt.dtype(torch::kFloat32));
+    acc = torch::empty({2, capacity}, devopt.dtype(torch::kFloat64));
+    outs = torch::empty({3, capacity}, devopt.dtype(torch::kFloat64));
+    hist = torch::empty({total_bins}, devopt.dtype(torch::kInt32));
+    ksd = torch::empty({N_NUM}, devopt.dtype(torch::kFloat32));
+
+    auto pinned = torch::TensorOptions().pinned_memory(true);
+    pin_codes = torch::empty({capacity, N_CAT}, pinned.dtype(torch::kInt16));
+    pin_nums = torch::empty({capacity, N_NUM}, pinned.dtype(torch::kFloat32));
+    pin_outs = torch::empty({3, capacity}, pinned.dtype(torch::kFloat64));
+    pin_hist = torch::empty({total_bins}, pinned.dtype(torch::kInt32));
+    pin_ksd = torch::empty({N_NUM}, pinned.dtype(torch::kFloat32));
+
+    HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+    HIP_CHECK(hipDeviceSynchronize());  // uploads above used torch's stream
+  }
+
+  ~ScoreSession() {
+    if (stream) {
+      (void)hipStreamSynchronize(stream);
+      (void)hipStreamDestroy(stream);
+    }
+  }
+
+  // Score b rows already staged in pin_codes/pin_nums. Blocks (GIL
+  // released) until pin_outs/pin_hist/pin_ksd hold the results.
+  void score(int64_t b64, bool with_drift, bool sync) {
+    TORCH_CHECK(b64 >= 1 && b64 <= capacity, "batch out of range: ", b64);
+    const int b = (int)b64;
+    py::gil_scoped_release nogil;
+    c10::hip::HIPGuard guard((c10::DeviceIndex)device_index);
+
+    HIP_CHECK(hipMemcpyAsync(d_codes.data_ptr(), pin_codes.data_ptr(),
+        (size_t)b * N_CAT * sizeof(short), hipMemcpyHostToDevice, stream));
+    HIP_CHECK(hipMemcpyAsync(d_nums.data_ptr(), pin_nums.data_ptr(),
+        (size_t)b * N_NUM * sizeof(float), hipMemcpyHostToDevice, stream));
+    double* acc_cls = acc.data_ptr<double>();
+    double* acc_if = acc_cls + capacity;
+    HIP_CHECK(hipMemsetAsync(acc_cls, 0, (size_t)b * sizeof(double), stream));
+    HIP_CHECK(hipMemsetAsync(acc_if, 0, (size_t)b * sizeof(double), stream));
+
+    const int row_blocks = ceil_div(b, BLOCK);
+    auto chunks = [&](int64_t t) {
+      return std::max(1, std::min(ceil_div(2048, row_blocks), (int)t));
+    };
+    hipLaunchKernelGGL((forest_kernel<false>), dim3(row_blocks, chunks(t_cls)),
+        dim3(BLOCK), 0, stream,
+        d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
+        reinterpret_cast<const int4*>(cls_nodes.data_ptr<int>()),
+        cls_off.data_ptr<int>(), (int)t_cls,
+        feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), b, acc_cls);
+    hipLaunchKernelGGL((forest_kernel<true>), dim3(row_blocks, chunks(t_if)),
+        dim3(BLOCK), 0, stream,
+        d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
+        reinterpret_cast<const int4*>(if_nodes.data_ptr<int>()),
+        if_off.data_ptr<int>(), (int)t_if, nullptr, nullptr, b, acc_if);
+
+    double* proba = outs.data_ptr<double>();
+    hipLaunchKernelGGL(finalize_kernel, dim3(row_blocks), dim3(BLOCK), 0, stream,
+        acc_cls, acc_if, b, 1.0 / (double)t_cls, if_denom, if_offset, if_threshold,
+        proba, proba + capacity, proba + 2 * capacity);
+
+    if (with_drift) {
+      HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0, (size_t)total_bins * sizeof(int), stream));
+      const int hist_blocks = std::min(row_blocks, 1024);
+      hipLaunchKernelGGL(cat_hist_kernel, dim3(hist_blocks), dim3(BLOCK),
+          (size_t)total_bins * sizeof(int), stream,
+          d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
+          hist.data_ptr<int>());
+      int m_pow2 = 2;
+      while (m_pow2 < b) m_pow2 <<= 1;
+      TORCH_CHECK(b <= MAX_DRIFT_ROWS, "drift batch too large: ", b);
+      hipLaunchKernelGGL(ks_kernel, dim3(N_NUM), dim3(BLOCK),
+          (size_t)m_pow2 * sizeof(float), stream,
+          d_nums.data_ptr<float>(), medians.data_ptr<float>(), b, m_pow2,
+          ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
+          ksd.data_ptr<float>());
+      HIP_CHECK(hipMemcpyAsync(pin_hist.data_ptr(), hist.data_ptr(),
+          (size_t)total_bins * sizeof(int), hipMemcpyDeviceToHost, stream));
+      HIP_CHECK(hipMemcpyAsync(pin_ksd.data_ptr(), ksd.data_ptr(),
+          (size_t)N_NUM * sizeof(float), hipMemcpyDeviceToHost, stream));
+    }
+    // proba/iscore/outlier rows are capacity-strided: copy row-by-row into
+    // a compact [3, b] pinned image (3 small async copies)
+    double* pout = pin_outs.data_ptr<double>();
+    for (int r = 0; r < 3; ++r)
+      HIP_CHECK(hipMemcpyAsync(pout + (size_t)r * capacity, proba + (size_t)r * capacity,
+          (size_t)b * sizeof(double), hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipGetLastError());
+    if (sync) HIP_CHECK(hipStreamSynchronize(stream));
+  }
+
+  void synchronize() {
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipStreamSynchronize(stream));
+  }
+};
+
+// ---------------------------------------------------------------------------
 // Host-side request encoder (the native data-loader for the serving path).
 // Replaces the reference's pandas DataFrame construction + sklearn
 // OneHotEncoder lookup per request (reference app/main.py:54 →
@@ -462,4 +607,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Per-feature drift statistics: categorical histograms + K-S D (gfx950)");
   m.def("encode_records", &encode_records,
         "Native request encoder: list[dict] -> (codes i16[B,9], nums f32[B,14])");
+  py::class_<ScoreSession>(m, "ScoreSession")
+      .def(py::init<py::dict, int64_t, int>(), py::arg("model"),
+           py::arg("capacity"), py::arg("device_index"))
+      .def("score", &ScoreSession::score, py::arg("b"),
+           py::arg("with_drift") = true, py::arg("sync") = true)
+      .def("synchronize", &ScoreSession::synchronize)
+      .def_readonly("capacity", &ScoreSession::capacity)
+      .def_property_readonly("pin_codes", [](ScoreSession& s) { return s.pin_codes; })
+      .def_property_readonly("pin_nums", [](ScoreSession& s) { return s.pin_nums; })
+      .def_property_readonly("pin_outs", [](ScoreSession& s) { return s.pin_outs; })
+      .def_property_readonly("pin_hist", [](ScoreSession& s) { return s.pin_hist; })
+      .def_property_readonly("pin_ksd", [](ScoreSession& s) { return s.pin_ksd; });
 }
