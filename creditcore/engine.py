@@ -103,10 +103,11 @@ class ScoringEngine:
         g["np_codes"][:b] = codes
         g["np_nums"][:b] = nums
         g["sess"].score(b, drift_now, True)  # blocks; GIL released
+        flat = g["np_outs"].reshape(-1)  # b-packed: proba | iscore | outlier
         out = {
-            "predictions": g["np_outs"][0, :b].copy(),
-            "instance_score": g["np_outs"][1, :b].copy(),
-            "outliers": g["np_outs"][2, :b].copy(),
+            "predictions": flat[:b].copy(),
+            "instance_score": flat[b : 2 * b].copy(),
+            "outliers": flat[2 * b : 3 * b].copy(),
         }
         if with_drift and not drift_now:
             # oversized batch: score a capped drift sample in a second pass

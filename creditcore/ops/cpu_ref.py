@@ -143,12 +143,25 @@ def pvals_from_stats(
     Vectorized: exactly two scipy sf calls per request."""
     from ..models.drift import chi2_from_counts_many, ks_asymp_pvalue_many
 
+    # ref columns all have n_ref rows (fitted on one matrix)
+    n_ref = int(packed.ref_sorted_offsets[1] - packed.ref_sorted_offsets[0])
+    en = round(n_ref * n_batch / (n_ref + n_batch))
+    if en >= 300:  # Pelz-Good accuracy region (see ks_asymp_pvalue_many)
+        from . import gpu
+
+        if gpu.available():  # native epilogue: ~5 µs vs ~0.4 ms in numpy
+            return gpu._ext.drift_pvals_host(
+                np.ascontiguousarray(cat_hists, dtype=np.int32),
+                np.ascontiguousarray(ks_d, dtype=np.float32),
+                packed.ref_cat_counts,
+                packed.ref_cat_offsets,
+                n_ref,
+                n_batch,
+            )
     pvals = np.ones(N_CAT + N_NUM, dtype=np.float64)
     pvals[:N_CAT] = chi2_from_counts_many(
         packed.ref_cat_counts, cat_hists, packed.ref_cat_offsets
     )
-    # ref columns all have n_ref rows (fitted on one matrix)
-    n_ref = int(packed.ref_sorted_offsets[1] - packed.ref_sorted_offsets[0])
     pvals[N_CAT:] = ks_asymp_pvalue_many(ks_d, n_ref, n_batch)
     return pvals
 
