@@ -105,16 +105,18 @@ def main():
 
     engine = ScoringEngine(packed, device=device, device_index=local_rank)
 
-    # Pre-generate a pool of request bodies (list-of-dict JSON shapes); each
-    # timed step scores one full request through the serving compute path.
+    # Pre-generate a pool of request bodies in the wire format (JSON bytes);
+    # each timed step runs the full serving compute path: native JSON parse
+    # + encode -> pinned H2D -> HIP kernels -> D2H -> drift p-values ->
+    # response assembly.
     pool = [
-        make_request_batch(args.rows, seed=100 + rank * 16 + i)
+        json.dumps(make_request_batch(args.rows, seed=100 + rank * 16 + i)).encode()
         for i in range(4)
     ]
     with_drift = not args.no_drift
 
     def one_step(i: int):
-        out = engine.score_records(pool[i % len(pool)])
+        out = engine.score_json(pool[i % len(pool)])
         assert out["rows"] == args.rows
         return out
 

@@ -138,10 +138,53 @@ class ScoringEngine:
             out["ks_d"] = ks_d
         return out
 
+    def score_json(self, body: bytes) -> dict:
+        """Score a raw /score JSON request body. Uses the native C parser
+        (no intermediate Python objects) when the extension is present;
+        raises ValueError on malformed/ill-typed bodies (callers fall back
+        to pydantic for proper 422 semantics)."""
+        from .ops import gpu
+        from .pack import CATEGORICAL_FEATURES, MISSING_CATEGORY, NUMERIC_FEATURES
+
+        if gpu.available():
+            dc, dn = self.default_rows()
+            codes, nums = gpu._ext.encode_json(
+                body,
+                self.packed.vocabs,
+                CATEGORICAL_FEATURES,
+                NUMERIC_FEATURES,
+                MISSING_CATEGORY,
+                dc,
+                dn,
+            )
+            codes = np.asarray(codes)
+            nums = np.asarray(nums)
+        else:
+            import json
+
+            codes, nums = encode_batch(json.loads(body), self.packed.vocabs)
+        return self._score_encoded(codes, nums)
+
+    def default_rows(self) -> tuple:
+        """Encoded schema-default record (absent request fields take these
+        values — pydantic default semantics, reference app/model.py:8-34)."""
+        if getattr(self, "_default_rows", None) is None:
+            from .schema import LoanApplicant
+
+            dc, dn = encode_batch([LoanApplicant().__dict__], self.packed.vocabs)
+            self._default_rows = (
+                np.ascontiguousarray(dc[0]),
+                np.ascontiguousarray(dn[0]),
+            )
+        return self._default_rows
+
     def score_records(self, records) -> dict:
         """Score a request body (list of dicts / DataFrame); returns the
         reference response shape (02-register cell-9)."""
         codes, nums = encode_batch(records, self.packed.vocabs)
+        return self._score_encoded(codes, nums)
+
+    def _score_encoded(self, codes: np.ndarray, nums: np.ndarray) -> dict:
         t0 = time.perf_counter()
         raw = self.score_arrays(codes, nums)
         latency_ms = (time.perf_counter() - t0) * 1e3

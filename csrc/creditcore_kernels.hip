@@ -650,6 +650,293 @@ py::tuple encode_records(py::list recs, py::list vocabs, py::list cat_names,
 }
 
 // ---------------------------------------------------------------------------
+// encode_json — parse a /score request body (JSON bytes, the wire format:
+// [{"sex": "male", ..., "credit_limit": 18000.0, ...}, ...]) straight into
+// (codes, nums) with no intermediate Python objects. This is the serving
+// fast path; any deviation from the expected shape throws and the caller
+// falls back to the pydantic path for a proper 422. The GIL is released
+// during the parse.
+// ---------------------------------------------------------------------------
+
+#include <charconv>
+#include <cstring>
+
+namespace jsonenc {
+
+struct Error {
+  std::string msg;
+};
+
+struct Parser {
+  const char* p;
+  const char* end;
+
+  [[noreturn]] void fail(const char* what) {
+    throw Error{std::string(what) + " at offset " + std::to_string((size_t)(p - begin_))};
+  }
+  const char* begin_;
+
+  void ws() {
+    while (p < end && (*p == ' ' || *p == '\t' || *p == '\n' || *p == '\r')) ++p;
+  }
+  bool eat(char c) {
+    ws();
+    if (p < end && *p == c) { ++p; return true; }
+    return false;
+  }
+  void expect(char c, const char* what) {
+    if (!eat(c)) fail(what);
+  }
+
+  // Parse a JSON string starting at the opening quote; returns the raw
+  // span between quotes and whether it contains escapes.
+  void raw_string(const char*& s, size_t& len, bool& escaped) {
+    expect('"', "expected string");
+    s = p;
+    escaped = false;
+    while (p < end) {
+      const char c = *p;
+      if (c == '"') { len = (size_t)(p - s); ++p; return; }
+      if (c == '\\') {
+        escaped = true;
+        ++p;
+        if (p >= end) break;
+      }
+      ++p;
+    }
+    fail("unterminated string");
+  }
+
+  // Unescape into buf (capacity cap); returns length or SIZE_MAX if too long.
+  size_t unescape(const char* s, size_t len, char* buf, size_t cap) {
+    size_t o = 0;
+    for (size_t i = 0; i < len; ++i) {
+      char c = s[i];
+      if (c != '\\') {
+        if (o >= cap) return SIZE_MAX;
+        buf[o++] = c;
+        continue;
+      }
+      if (++i >= len) return SIZE_MAX;
+      c = s[i];
+      char repl;
+      switch (c) {
+        case '"': repl = '"'; break;
+        case '\\': repl = '\\'; break;
+        case '/': repl = '/'; break;
+        case 'b': repl = '\b'; break;
+        case 'f': repl = '\f'; break;
+        case 'n': repl = '\n'; break;
+        case 'r': repl = '\r'; break;
+        case 't': repl = '\t'; break;
+        case 'u': {
+          if (i + 4 >= len) return SIZE_MAX;
+          unsigned cp = 0;
+          for (int k = 1; k <= 4; ++k) {
+            const char h = s[i + k];
+            cp <<= 4;
+            if (h >= '0' && h <= '9') cp |= (unsigned)(h - '0');
+            else if (h >= 'a' && h <= 'f') cp |= (unsigned)(h - 'a' + 10);
+            else if (h >= 'A' && h <= 'F') cp |= (unsigned)(h - 'A' + 10);
+            else return SIZE_MAX;
+          }
+          i += 4;
+          // UTF-8 encode (surrogate pairs unsupported here: vocab is ASCII,
+          // a non-matching value only needs to be *skipped* correctly)
+          if (cp < 0x80) {
+            if (o >= cap) return SIZE_MAX;
+            buf[o++] = (char)cp;
+          } else if (cp < 0x800) {
+            if (o + 2 > cap) return SIZE_MAX;
+            buf[o++] = (char)(0xC0 | (cp >> 6));
+            buf[o++] = (char)(0x80 | (cp & 0x3F));
+          } else {
+            if (o + 3 > cap) return SIZE_MAX;
+            buf[o++] = (char)(0xE0 | (cp >> 12));
+            buf[o++] = (char)(0x80 | ((cp >> 6) & 0x3F));
+            buf[o++] = (char)(0x80 | (cp & 0x3F));
+          }
+          continue;
+        }
+        default: return SIZE_MAX;
+      }
+      if (o >= cap) return SIZE_MAX;
+      buf[o++] = repl;
+    }
+    return o;
+  }
+
+  double number() {
+    ws();
+    double v;
+    auto [q, ec] = std::from_chars(p, end, v);
+    if (ec != std::errc()) fail("expected number");
+    p = q;
+    return v;
+  }
+
+  bool literal(const char* lit) {
+    const size_t n = std::strlen(lit);
+    if ((size_t)(end - p) >= n && std::memcmp(p, lit, n) == 0) { p += n; return true; }
+    return false;
+  }
+
+  void skip_value() {
+    ws();
+    if (p >= end) fail("truncated value");
+    const char c = *p;
+    if (c == '"') {
+      const char* s; size_t l; bool e;
+      raw_string(s, l, e);
+    } else if (c == '{' || c == '[') {
+      const char open = c, close = (c == '{') ? '}' : ']';
+      int depth = 0;
+      while (p < end) {
+        const char d = *p;
+        if (d == '"') { const char* s; size_t l; bool e; raw_string(s, l, e); continue; }
+        if (d == open) ++depth;
+        else if (d == close && --depth == 0) { ++p; return; }
+        ++p;
+      }
+      fail("unterminated container");
+    } else if (literal("null") || literal("true") || literal("false")) {
+    } else {
+      number();
+    }
+  }
+};
+
+}  // namespace jsonenc
+
+py::tuple encode_json(py::bytes body, py::list vocabs, py::list cat_names,
+                      py::list num_names, py::str missing_cat,
+                      py::array_t<int16_t> default_codes,
+                      py::array_t<float> default_nums) {
+  const int ncat = (int)py::len(cat_names);
+  const int nnum = (int)py::len(num_names);
+  // column tables: name -> id (cats: [0, ncat), nums: [ncat, ncat+nnum));
+  // string_view keys into stable storage: no allocation per field lookup
+  std::vector<std::string> key_store;
+  key_store.reserve(ncat + nnum);
+  std::unordered_map<std::string_view, int> colmap;
+  std::vector<std::vector<std::string>> vocab(ncat);
+  std::vector<int16_t> missing_code(ncat, -1);
+  const std::string miss = py::cast<std::string>(missing_cat);
+  for (int j = 0; j < ncat; ++j) {
+    key_store.push_back(py::cast<std::string>(cat_names[j]));
+    py::list v = vocabs[j];
+    for (ssize_t k = 0; k < py::len(v); ++k) {
+      vocab[j].push_back(py::cast<std::string>(v[k]));
+      if (vocab[j].back() == miss) missing_code[j] = (int16_t)k;
+    }
+  }
+  for (int j = 0; j < nnum; ++j)
+    key_store.push_back(py::cast<std::string>(num_names[j]));
+  for (int j = 0; j < ncat + nnum; ++j) colmap[key_store[j]] = j;
+
+  TORCH_CHECK(default_codes.size() == ncat && default_nums.size() == nnum,
+      "default row size mismatch");
+  const int16_t* def_codes = default_codes.data();
+  const float* def_nums = default_nums.data();
+  char* data;
+  ssize_t blen;
+  if (PyBytes_AsStringAndSize(body.ptr(), &data, &blen) != 0)
+    throw py::value_error("body must be bytes");
+
+  std::vector<int16_t> codes;
+  std::vector<float> nums;
+  size_t b = 0;
+  {
+    py::gil_scoped_release nogil;
+    jsonenc::Parser P{data, data + blen};
+    P.begin_ = data;
+    codes.reserve((size_t)blen / 70 * ncat + ncat);
+    nums.reserve((size_t)blen / 70 * nnum + nnum);
+    try {
+      P.expect('[', "body must be a JSON array");
+      if (!P.eat(']')) {
+        do {
+          P.expect('{', "record must be an object");
+          codes.resize(codes.size() + ncat);
+          nums.resize(nums.size() + nnum);
+          int16_t* crow = codes.data() + b * ncat;
+          float* nrow = nums.data() + b * nnum;
+          // absent fields take the schema defaults (pydantic semantics,
+          // reference app/model.py:8-34)
+          std::memcpy(crow, def_codes, ncat * sizeof(int16_t));
+          std::memcpy(nrow, def_nums, nnum * sizeof(float));
+          if (!P.eat('}')) {
+            do {
+              const char* ks; size_t kl; bool kesc;
+              P.raw_string(ks, kl, kesc);
+              char kbuf[64];
+              if (kesc) {
+                const size_t n = P.unescape(ks, kl, kbuf, sizeof(kbuf));
+                if (n == SIZE_MAX) P.fail("bad key");
+                ks = kbuf; kl = n;
+              }
+              P.expect(':', "expected ':'");
+              auto it = colmap.find(std::string_view(ks, kl));
+              if (it == colmap.end()) {
+                P.skip_value();  // extra fields ignored (schema extra="ignore")
+              } else if (it->second < ncat) {
+                const int j = it->second;
+                P.ws();
+                if (P.p < P.end && *P.p == 'n') {
+                  P.fail("null not accepted for a string field");
+                } else {
+                  const char* vs; size_t vl; bool vesc;
+                  P.raw_string(vs, vl, vesc);
+                  char vbuf[64];
+                  if (vesc) {
+                    const size_t n = P.unescape(vs, vl, vbuf, sizeof(vbuf));
+                    if (n == SIZE_MAX) { crow[j] = -1; continue; }
+                    vs = vbuf; vl = n;
+                  }
+                  int16_t code = -1;  // unknown category -> all-zero one-hot
+                  for (size_t k = 0; k < vocab[j].size(); ++k) {
+                    const std::string& cand = vocab[j][k];
+                    if (cand.size() == vl && std::memcmp(cand.data(), vs, vl) == 0) {
+                      code = (int16_t)k;
+                      break;
+                    }
+                  }
+                  crow[j] = code;
+                }
+              } else {
+                const int j = it->second - ncat;
+                P.ws();
+                if (P.p < P.end && *P.p == 'n') {
+                  P.fail("null not accepted for a numeric field");
+                } else {
+                  nrow[j] = (float)P.number();
+                }
+              }
+            } while (P.eat(','));
+            P.expect('}', "expected '}'");
+          }
+          ++b;
+        } while (P.eat(','));
+        P.expect(']', "expected ']'");
+      }
+      P.ws();
+      if (P.p != P.end) P.fail("trailing data");
+    } catch (const jsonenc::Error& e) {
+      py::gil_scoped_acquire gil;
+      throw py::value_error(e.msg);
+    }
+  }
+
+  py::array_t<int16_t> codes_arr({(ssize_t)b, (ssize_t)ncat});
+  py::array_t<float> nums_arr({(ssize_t)b, (ssize_t)nnum});
+  if (b) {
+    std::memcpy(codes_arr.mutable_data(), codes.data(), b * ncat * sizeof(int16_t));
+    std::memcpy(nums_arr.mutable_data(), nums.data(), b * nnum * sizeof(float));
+  }
+  return py::make_tuple(codes_arr, nums_arr);
+}
+
+// ---------------------------------------------------------------------------
 // Drift p-value epilogue on host, in C: chi-square survival via the exact
 // closed forms for integer dof (even: Poisson tail sum; odd: erfc +
 // half-integer-gamma series, both from the Q(k+2) = Q(k) + term recurrence)
@@ -790,6 +1077,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Per-feature drift statistics: categorical histograms + K-S D (gfx950)");
   m.def("encode_records", &encode_records,
         "Native request encoder: list[dict] -> (codes i16[B,9], nums f32[B,14])");
+  m.def("encode_json", &encode_json,
+        "Parse a /score JSON request body straight into (codes, nums)");
   m.def("drift_pvals_host", &drift_pvals_host,
         "Drift p-values from kernel statistics (chi2 + Pelz-Good K-S), host C");
   py::class_<ScoreSession>(m, "ScoreSession")

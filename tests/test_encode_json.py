@@ -1,0 +1,121 @@
+"""Native JSON request parser (csrc encode_json) vs the pydantic reference
+path — the serving fast path must be byte-for-byte equivalent on valid
+bodies and must *reject* (→ fallback) anything pydantic would treat
+differently (nulls, wrong types)."""
+
+from __future__ import annotations
+
+import json
+
+import numpy as np
+import pytest
+
+from creditcore.pack import encode_batch
+from creditcore.schema import (
+    CATEGORICAL_FEATURES,
+    MISSING_CATEGORY,
+    NUMERIC_FEATURES,
+    LoanApplicant,
+)
+
+
+@pytest.fixture(scope="module")
+def ext():
+    import torch  # noqa: F401  (loads libc10 for the extension)
+
+    from creditcore.ops import gpu
+
+    if not gpu.available():
+        pytest.skip("extension not built")
+    return gpu.ext()
+
+
+@pytest.fixture(scope="module")
+def defaults(packed):
+    dc, dn = encode_batch([LoanApplicant().__dict__], packed.vocabs)
+    return np.ascontiguousarray(dc[0]), np.ascontiguousarray(dn[0])
+
+
+def _enc(ext, packed, defaults, body: bytes):
+    c, n = ext.encode_json(
+        body, packed.vocabs, CATEGORICAL_FEATURES, NUMERIC_FEATURES,
+        MISSING_CATEGORY, defaults[0], defaults[1],
+    )
+    return np.asarray(c), np.asarray(n)
+
+
+def test_parity_with_pydantic_path(ext, packed, defaults):
+    from creditcore.data import make_request_batch
+
+    recs = make_request_batch(257, seed=3)
+    body = json.dumps(recs).encode()
+    c1, n1 = _enc(ext, packed, defaults, body)
+    validated = [LoanApplicant(**r).__dict__ for r in recs]
+    c2, n2 = encode_batch(validated, packed.vocabs)
+    np.testing.assert_array_equal(c1, c2)
+    np.testing.assert_array_equal(n1, n2)
+
+
+def test_absent_fields_take_schema_defaults(ext, packed, defaults):
+    c, n = _enc(ext, packed, defaults, b"[{}]")
+    validated = [LoanApplicant().__dict__]
+    c2, n2 = encode_batch(validated, packed.vocabs)
+    np.testing.assert_array_equal(c, c2)
+    np.testing.assert_array_equal(n, n2)
+
+
+def test_unknown_category_and_extra_fields(ext, packed, defaults):
+    body = b'[{"education": "zzz", "extra": {"deep": [1, "x", {}]}, "age": 44}]'
+    c, n = _enc(ext, packed, defaults, body)
+    assert c[0, CATEGORICAL_FEATURES.index("education")] == -1
+    assert n[0, NUMERIC_FEATURES.index("age")] == 44.0
+
+
+def test_escapes_and_numbers(ext, packed, defaults):
+    body = b'[{"sex": "mal\\u0065", "credit_limit": 1.5e4, "age": -3}]'
+    c, n = _enc(ext, packed, defaults, body)
+    assert c[0, 0] == packed.vocabs[0].index("male")
+    assert n[0, 0] == 15000.0
+    assert n[0, 1] == -3.0
+
+
+@pytest.mark.parametrize(
+    "bad",
+    [
+        b"{}",  # not an array
+        b'[{"sex": 5}]',  # number for string field
+        b'[{"sex": null}]',  # null (pydantic rejects -> 422 via fallback)
+        b'[{"credit_limit": "x"}]',  # string for numeric field
+        b'[{"credit_limit": null}]',
+        b"[",  # truncated
+        b'[{"sex" "male"}]',  # missing colon
+        b'[{"sex": "male"}] trailing',
+    ],
+)
+def test_malformed_rejected(ext, packed, defaults, bad):
+    with pytest.raises(ValueError):
+        _enc(ext, packed, defaults, bad)
+
+
+def test_empty_array(ext, packed, defaults):
+    c, n = _enc(ext, packed, defaults, b" [ ] ")
+    assert c.shape == (0, 9) and n.shape == (0, 14)
+
+
+def test_serving_fallback_handles_lax_types(model_dir):
+    """A numeric string coerces through the pydantic fallback exactly as the
+    reference would (fast path rejects, fallback accepts)."""
+    from fastapi.testclient import TestClient
+
+    from creditcore.config import ServeConfig
+    from creditcore.serve import create_app
+
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    app = create_app(cfg)
+    with TestClient(app) as client:
+        r = client.post("/predict", json=[{"credit_limit": "18000"}])
+        assert r.status_code == 200
+        r = client.post("/predict", json=[{"sex": None}])
+        assert r.status_code == 422
