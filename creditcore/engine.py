@@ -138,31 +138,36 @@ class ScoringEngine:
             out["ks_d"] = ks_d
         return out
 
-    def score_json(self, body: bytes) -> dict:
-        """Score a raw /score JSON request body. Uses the native C parser
-        (no intermediate Python objects) when the extension is present;
-        raises ValueError on malformed/ill-typed bodies (callers fall back
-        to pydantic for proper 422 semantics)."""
+    def encode_json_body(self, body: bytes) -> tuple:
+        """Parse a raw /score JSON body into (codes, nums) with the native C
+        parser (GIL released during the parse); raises ValueError on
+        malformed/ill-typed bodies (callers fall back to pydantic for proper
+        422 semantics)."""
         from .ops import gpu
         from .pack import CATEGORICAL_FEATURES, MISSING_CATEGORY, NUMERIC_FEATURES
 
         if gpu.available():
-            dc, dn = self.default_rows()
-            codes, nums = gpu._ext.encode_json(
-                body,
-                self.packed.vocabs,
-                CATEGORICAL_FEATURES,
-                NUMERIC_FEATURES,
-                MISSING_CATEGORY,
-                dc,
-                dn,
-            )
-            codes = np.asarray(codes)
-            nums = np.asarray(nums)
-        else:
-            import json
+            enc = getattr(self, "_json_encoder", None)
+            if enc is None:
+                dc, dn = self.default_rows()
+                enc = gpu.ext().JsonEncoder(
+                    self.packed.vocabs,
+                    CATEGORICAL_FEATURES,
+                    NUMERIC_FEATURES,
+                    MISSING_CATEGORY,
+                    dc,
+                    dn,
+                )
+                self._json_encoder = enc
+            codes, nums = enc.encode(body)
+            return np.asarray(codes), np.asarray(nums)
+        import json
 
-            codes, nums = encode_batch(json.loads(body), self.packed.vocabs)
+        return encode_batch(json.loads(body), self.packed.vocabs)
+
+    def score_json(self, body: bytes) -> dict:
+        """Score a raw /score JSON request body (wire-format fast path)."""
+        codes, nums = self.encode_json_body(body)
         return self._score_encoded(codes, nums)
 
     def default_rows(self) -> tuple:

@@ -147,13 +147,7 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         vocabs = engine.packed.vocabs
         if gpu.available():
             try:
-                ext = gpu.ext()
-                dc, dn = engine.default_rows()
-                codes, nums = ext.encode_json(
-                    body, vocabs, CATEGORICAL_FEATURES, NUMERIC_FEATURES,
-                    MISSING_CATEGORY, dc, dn,
-                )
-                return np.asarray(codes), np.asarray(nums)
+                return engine.encode_json_body(body)
             except ValueError:
                 pass  # fall through to full validation
         from pydantic import TypeAdapter, ValidationError
