@@ -87,8 +87,72 @@ def _cmd_smoke(argv):
     print("SMOKE OK")
 
 
+def _cmd_pipeline(argv):
+    """The full local CD pipeline (reference deploy workflow analog)."""
+    p = argparse.ArgumentParser(prog="creditcore pipeline")
+    p.add_argument("--model-dir", default="./model")
+    p.add_argument("--registry-root", default=None)
+    p.add_argument("--model-name", default="credit-default-uci-custom")
+    p.add_argument("--max-evals", type=int, default=10)
+    p.add_argument("--n-rows", type=int, default=20_000)
+    p.add_argument("--staging-port", type=int, default=5001)
+    p.add_argument("--production-port", type=int, default=5000)
+    p.add_argument("--device", default="auto")
+    p.add_argument("--auto-approve", action="store_true")
+    p.add_argument("--keep-production", action="store_true")
+    p.add_argument("--docker-image", default=None)
+    a = p.parse_args(argv)
+    from .pipeline import run_pipeline
+
+    report = run_pipeline(
+        model_dir=a.model_dir,
+        registry_root=a.registry_root,
+        model_name=a.model_name,
+        max_evals=a.max_evals,
+        n_rows=a.n_rows,
+        staging_port=a.staging_port,
+        production_port=a.production_port,
+        device=a.device,
+        auto_approve=a.auto_approve,
+        keep_production=a.keep_production,
+        docker_image=a.docker_image,
+    )
+    print(json.dumps(report, indent=2))
+
+
+def _cmd_generate_data(argv):
+    """Write a UCI-shaped synthetic CSV (the reference's curated.csv analog,
+    reference databricks/data/; the real UCI CSV is not available offline)."""
+    p = argparse.ArgumentParser(prog="creditcore generate-data")
+    p.add_argument("--out", default="./data/curated.csv")
+    p.add_argument("--n-rows", type=int, default=30_000)
+    p.add_argument("--seed", type=int, default=2024)
+    p.add_argument("--inference-sample", default=None,
+                   help="also write an N-row inference.csv-style sample")
+    a = p.parse_args(argv)
+    import os
+
+    from .data import make_uci_shaped_frame
+
+    os.makedirs(os.path.dirname(a.out) or ".", exist_ok=True)
+    df = make_uci_shaped_frame(n_rows=a.n_rows, seed=a.seed)
+    df.to_csv(a.out, index=False)
+    print(f"wrote {a.out} ({len(df)} rows)")
+    if a.inference_sample:
+        df2 = make_uci_shaped_frame(n_rows=80, seed=a.seed + 1, include_target=False)
+        df2.to_csv(a.inference_sample, index=False)
+        print(f"wrote {a.inference_sample} (80 rows)")
+
+
 def main():
-    cmds = {"train": _cmd_train, "pack": _cmd_pack, "serve": _cmd_serve, "smoke": _cmd_smoke}
+    cmds = {
+        "train": _cmd_train,
+        "pack": _cmd_pack,
+        "serve": _cmd_serve,
+        "smoke": _cmd_smoke,
+        "pipeline": _cmd_pipeline,
+        "generate-data": _cmd_generate_data,
+    }
     if len(sys.argv) < 2 or sys.argv[1] not in cmds:
         print(f"usage: python -m creditcore {{{'|'.join(cmds)}}} [args]", file=sys.stderr)
         sys.exit(2)
