@@ -403,6 +403,8 @@ struct ScoreSession {
   int64_t total_bins{}, t_cls{}, t_if{}, capacity{};
   int device_index{};
   hipStream_t stream{};
+  hipStream_t stream2{};  // drift branch (runs parallel to the forests)
+  hipEvent_t ev_fork{}, ev_join{};
 
   ScoreSession(py::dict model, int64_t cap, int dev) : capacity(cap), device_index(dev) {
     c10::hip::HIPGuard guard((c10::DeviceIndex)dev);
@@ -454,6 +456,9 @@ struct ScoreSession {
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)KS_LDS_BYTES);
 
     HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream2, hipStreamNonBlocking));
+    HIP_CHECK(hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&ev_join, hipEventDisableTiming));
     HIP_CHECK(hipDeviceSynchronize());  // uploads above used torch's stream
   }
 
@@ -462,6 +467,9 @@ struct ScoreSession {
       (void)hipStreamSynchronize(stream);
       for (auto& kv : graphs) (void)hipGraphExecDestroy(kv.second);
       (void)hipStreamDestroy(stream);
+      (void)hipStreamDestroy(stream2);
+      (void)hipEventDestroy(ev_fork);
+      (void)hipEventDestroy(ev_join);
     }
   }
 
@@ -473,6 +481,11 @@ struct ScoreSession {
         (size_t)b * N_CAT * sizeof(short), hipMemcpyHostToDevice, stream));
     HIP_CHECK(hipMemcpyAsync(d_nums.data_ptr(), pin_nums.data_ptr(),
         (size_t)b * N_NUM * sizeof(float), hipMemcpyHostToDevice, stream));
+    // fork point: the drift branch (stream2) depends only on the H2D copies
+    if (with_drift) {
+      HIP_CHECK(hipEventRecord(ev_fork, stream));
+      HIP_CHECK(hipStreamWaitEvent(stream2, ev_fork, 0));
+    }
     double* acc_cls = acc.data_ptr<double>();
     double* acc_if = acc_cls + b;  // b-packed: one memset clears both
     HIP_CHECK(hipMemsetAsync(acc_cls, 0, (size_t)(2 * b) * sizeof(double), stream));
@@ -499,10 +512,13 @@ struct ScoreSession {
         proba, proba + b, proba + 2 * b);
 
     if (with_drift) {
-      HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0, (size_t)total_bins * sizeof(int), stream));
+      // Drift branch forked onto stream2 right after the H2D copies: the
+      // K-S sort+scan and categorical histogram overlap the forest kernels
+      // (captured as parallel graph branches; joined before the end).
+      HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0, (size_t)total_bins * sizeof(int), stream2));
       const int hist_blocks = std::min(row_blocks, 1024);
       hipLaunchKernelGGL(cat_hist_kernel, dim3(hist_blocks), dim3(BLOCK),
-          (size_t)total_bins * sizeof(int), stream,
+          (size_t)total_bins * sizeof(int), stream2,
           d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
           hist.data_ptr<int>());
       int m_pow2 = 2;
@@ -513,14 +529,16 @@ struct ScoreSession {
       const size_t ref_bytes = (size_t)max_ref_len * sizeof(float);
       const bool ref_lds = batch_bytes + ref_bytes <= KS_LDS_BYTES;
       hipLaunchKernelGGL(ks_kernel, dim3(N_NUM), dim3(BLOCK),
-          batch_bytes + (ref_lds ? ref_bytes : 0), stream,
+          batch_bytes + (ref_lds ? ref_bytes : 0), stream2,
           d_nums.data_ptr<float>(), medians.data_ptr<float>(), b, m_pow2,
           (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
           ksd.data_ptr<float>());
       HIP_CHECK(hipMemcpyAsync(pin_hist.data_ptr(), hist.data_ptr(),
-          (size_t)total_bins * sizeof(int), hipMemcpyDeviceToHost, stream));
+          (size_t)total_bins * sizeof(int), hipMemcpyDeviceToHost, stream2));
       HIP_CHECK(hipMemcpyAsync(pin_ksd.data_ptr(), ksd.data_ptr(),
-          (size_t)N_NUM * sizeof(float), hipMemcpyDeviceToHost, stream));
+          (size_t)N_NUM * sizeof(float), hipMemcpyDeviceToHost, stream2));
+      HIP_CHECK(hipEventRecord(ev_join, stream2));
+      HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
     }
     HIP_CHECK(hipMemcpyAsync(pin_outs.data_ptr(), proba,
         (size_t)(3 * b) * sizeof(double), hipMemcpyDeviceToHost, stream));
