@@ -1,0 +1,81 @@
+"""HTTP load generator for the serving stack.
+
+Closed-loop workers POST wire-format /score requests against a running
+server and report whole-service requests/sec + latency percentiles —
+the serving-layer view of the headline metric (bench.py measures the
+engine path; this measures through uvicorn/HTTP).
+
+    python -m creditcore serve --port 5000 &
+    python bench/load_gen.py --url http://127.0.0.1:5000 \
+        --rows 1024 --concurrency 8 --duration 30
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import time
+
+import numpy as np
+
+
+async def worker(client, url, body, stop_at, stats):
+    while time.perf_counter() < stop_at:
+        t0 = time.perf_counter()
+        r = await client.post(f"{url}/score", content=body,
+                              headers={"content-type": "application/json"})
+        dt = (time.perf_counter() - t0) * 1e3
+        if r.status_code == 200:
+            stats["lat"].append(dt)
+        else:
+            stats["errors"] += 1
+
+
+async def run(args):
+    import httpx
+
+    from creditcore.data import make_request_batch
+
+    body = json.dumps(make_request_batch(args.rows, seed=1)).encode()
+    stats = {"lat": [], "errors": 0}
+    async with httpx.AsyncClient(timeout=60.0) as client:
+        # warmup
+        await client.post(f"{args.url}/score", content=body,
+                          headers={"content-type": "application/json"})
+        stop_at = time.perf_counter() + args.duration
+        t0 = time.perf_counter()
+        await asyncio.gather(
+            *(worker(client, args.url, body, stop_at, stats)
+              for _ in range(args.concurrency))
+        )
+        elapsed = time.perf_counter() - t0
+    lat = np.sort(np.asarray(stats["lat"]))
+    n = len(lat)
+    out = {
+        "metric": "HTTP requests/sec, credit-default /score",
+        "value": round(n / elapsed, 2),
+        "unit": "requests/s",
+        "rows_per_request": args.rows,
+        "rows_per_sec": round(n * args.rows / elapsed, 1),
+        "concurrency": args.concurrency,
+        "duration_s": round(elapsed, 1),
+        "errors": stats["errors"],
+        "latency_ms_p50": round(float(lat[n // 2]), 2) if n else None,
+        "latency_ms_p90": round(float(lat[int(0.9 * n)]), 2) if n else None,
+        "latency_ms_p99": round(float(lat[min(n - 1, int(0.99 * n))]), 2) if n else None,
+    }
+    print(json.dumps(out))
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--url", default="http://127.0.0.1:5000")
+    p.add_argument("--rows", type=int, default=1024)
+    p.add_argument("--concurrency", type=int, default=8)
+    p.add_argument("--duration", type=float, default=20.0)
+    asyncio.run(run(p.parse_args()))
+
+
+if __name__ == "__main__":
+    main()

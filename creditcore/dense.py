@@ -1,0 +1,306 @@
+"""Large dense-tabular model family (BASELINE.json config 5:
+"Large synthetic tabular (10M rows × 1k feats) train + serve, 288 GB HBM
+sizing").
+
+The reference has no large-scale path at all — its training is a 4-vCPU
+sklearn job (reference train_register_model.yml:16-18) and its serving holds
+the whole drift reference inside a pickle (02-register cell-6). This family
+re-expresses the same three capabilities (classifier score, outlier flag,
+per-feature drift) at MI355X scale:
+
+- **train**: logistic regression on a streamed synthetic design matrix —
+  minibatch gradient descent on-GPU; the GEMMs go through rocBLAS (the
+  sanctioned library-GEMM path; everything non-GEMM here is fused HIP).
+- **score**: one fused HIP kernel (csrc dense_score_kernel): median
+  imputation + w·x + sigmoid + robust-z outlier per row, one wavefront per
+  row with coalesced feature loads.
+- **drift**: exact per-feature two-sample K-S against a sorted reference
+  column **resident in HBM** — at 10M rows × 1k features × f32 that is
+  40 GB per GPU, the 288 GB HBM3E sizing story; the generic ks_kernel runs
+  one block per feature (1k blocks fill the 256 CUs).
+"""
+
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+
+import numpy as np
+
+
+@dataclass
+class DenseModel:
+    """Packed dense model: flat arrays, mirror of pack.PackedModel for the
+    wide family."""
+
+    weight: np.ndarray  # f32[F]
+    bias: float
+    medians: np.ndarray  # f32[F]
+    inv_scale: np.ndarray  # f32[F] (1/IQR robust scale for the z outlier)
+    z_threshold: float
+    # drift reference (may be huge — kept as a memmap-able npy on disk)
+    ref_sorted: np.ndarray  # f32[F, n_ref] (per-feature sorted)
+    n_ref: int
+
+    @property
+    def n_features(self) -> int:
+        return int(len(self.weight))
+
+    def as_numpy(self) -> "DenseModel":
+        """Host copy (device-tensor-backed models from on-GPU training)."""
+        import torch
+
+        def np_(a):
+            return a.cpu().numpy() if torch.is_tensor(a) else a
+
+        return DenseModel(
+            weight=np_(self.weight),
+            bias=self.bias,
+            medians=np_(self.medians),
+            inv_scale=np_(self.inv_scale),
+            z_threshold=self.z_threshold,
+            ref_sorted=np_(self.ref_sorted),
+            n_ref=self.n_ref,
+        )
+
+    def save(self, path: str) -> None:
+        os.makedirs(path, exist_ok=True)
+        np.savez(
+            os.path.join(path, "dense_model.npz"),
+            weight=self.weight,
+            bias=self.bias,
+            medians=self.medians,
+            inv_scale=self.inv_scale,
+            z_threshold=self.z_threshold,
+            n_ref=self.n_ref,
+        )
+        # reference stored separately so it can be memmapped at load
+        np.save(os.path.join(path, "dense_ref_sorted.npy"), self.ref_sorted)
+
+    @classmethod
+    def load(cls, path: str, mmap: bool = True) -> "DenseModel":
+        z = np.load(os.path.join(path, "dense_model.npz"))
+        ref = np.load(
+            os.path.join(path, "dense_ref_sorted.npy"),
+            mmap_mode="r" if mmap else None,
+        )
+        return cls(
+            weight=z["weight"],
+            bias=float(z["bias"]),
+            medians=z["medians"],
+            inv_scale=z["inv_scale"],
+            z_threshold=float(z["z_threshold"]),
+            ref_sorted=ref,
+            n_ref=int(z["n_ref"]),
+        )
+
+
+def train_dense(
+    n_rows: int = 10_000_000,
+    n_feats: int = 1000,
+    ref_rows: int | None = None,
+    epochs: int = 1,
+    batch_rows: int = 65536,
+    lr: float = 0.5,
+    seed: int = 0,
+    device: str = "auto",
+    z_threshold: float = 6.0,
+    keep_on_device: bool = False,
+    log=print,
+) -> DenseModel:
+    """Train the dense family on streamed synthetic data.
+
+    Data is generated on-device in minibatches (there is no 40 GB dataset on
+    disk — the generator IS the dataset, seeded per-batch for
+    reproducibility). The drift reference is the per-feature sorted values
+    of ``ref_rows`` generated rows (defaults to min(n_rows, 2**21) on CPU
+    builds; pass the full n_rows on a GPU box to exercise the HBM sizing).
+    """
+    import torch
+
+    if device == "auto":
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+    dev = torch.device(device)
+    gen = torch.Generator(device=device).manual_seed(seed)
+
+    # ground-truth weights for the synthetic task (recoverable signal)
+    w_true = torch.randn(n_feats, generator=gen, device=dev) / (n_feats**0.5)
+
+    w = torch.zeros(n_feats, device=dev)
+    b = torch.zeros((), device=dev)
+    n_batches = max(1, n_rows // batch_rows)
+    for epoch in range(epochs):
+        for i in range(n_batches):
+            x = torch.randn(batch_rows, n_feats, generator=gen, device=dev)
+            logits_true = x @ w_true
+            y = (logits_true + 0.5 * torch.randn(batch_rows, generator=gen, device=dev) > 0).float()
+            # logistic regression step (GEMV/GEMM via rocBLAS)
+            z = x @ w + b
+            p = torch.sigmoid(z)
+            g = p - y
+            gw = x.t() @ g / batch_rows
+            gb = g.mean()
+            w -= lr * gw
+            b -= lr * gb
+            if i % 50 == 0:
+                with torch.no_grad():
+                    acc = ((p > 0.5).float() == y).float().mean().item()
+                log(f"[dense-train] epoch {epoch} batch {i}/{n_batches} acc={acc:.3f}")
+
+    # robust stats + drift reference from a seeded reference stream.
+    # Built in feature chunks so the transient peak stays bounded even for
+    # the 10M x 1k (40 GB) configuration: chunk randn + column sort into the
+    # preallocated [F, n_ref] buffer.
+    if ref_rows is None:
+        ref_rows = min(n_rows, 1 << 21)
+    rgen = torch.Generator(device=device).manual_seed(seed + 1)
+    ref_sorted = torch.empty(n_feats, ref_rows, device=dev, dtype=torch.float32)
+    med = torch.empty(n_feats, device=dev)
+    iqr = torch.empty(n_feats, device=dev)
+    chunk = max(1, min(n_feats, (1 << 28) // max(ref_rows, 1)))  # ≤ ~1 GB/chunk
+    for f0 in range(0, n_feats, chunk):
+        f1 = min(f0 + chunk, n_feats)
+        r = torch.randn(ref_rows, f1 - f0, generator=rgen, device=dev)
+        s = r.sort(dim=0).values  # [n_ref, chunk]
+        ref_sorted[f0:f1] = s.t()
+        med[f0:f1] = s[ref_rows // 2]
+        q1 = s[int(0.25 * (ref_rows - 1))]
+        q3 = s[int(0.75 * (ref_rows - 1))]
+        iqr[f0:f1] = (q3 - q1).clamp_min(1e-6)
+        del r, s
+
+    model = DenseModel(
+        weight=w.float(),
+        bias=float(b.item()),
+        medians=med.float(),
+        inv_scale=(1.0 / iqr).float(),
+        z_threshold=float(z_threshold),
+        ref_sorted=ref_sorted.float(),
+        n_ref=int(ref_rows),
+    )
+    if keep_on_device and device == "cuda":
+        # the 288 GB sizing path: the [F, n_ref] reference stays in HBM,
+        # DenseEngine consumes the tensors directly (no 40 GB host round trip)
+        return model
+    return model.as_numpy()
+
+
+class DenseEngine:
+    """Scoring engine for the dense family (one replica / GPU).
+
+    GPU path: fused dense_score_kernel + generic ks_kernel drift with the
+    sorted reference resident in HBM. CPU path: numpy reference with
+    identical semantics (golden tests compare the two).
+    """
+
+    MAX_DRIFT_ROWS = 16384
+
+    def __init__(self, model: DenseModel, device: str = "cpu", device_index: int = 0):
+        self.model = model
+        self.device = device
+        self.device_index = device_index
+        self._gpu = None
+        if device == "cuda":
+            self._init_gpu()
+
+    def _init_gpu(self):
+        import torch
+
+        from .ops import gpu
+
+        ext = gpu.ext()
+        dev = torch.device("cuda", self.device_index)
+        m = self.model
+
+        def up(a, dtype=torch.float32):
+            # model fields may already be device tensors (large-config train
+            # keeps the 40 GB reference on-GPU, no host round trip)
+            if torch.is_tensor(a):
+                return a.to(dev, dtype)
+            return torch.from_numpy(np.ascontiguousarray(a)).to(dev, dtype)
+
+        F = m.n_features
+        rs_off = np.arange(F + 1, dtype=np.int32) * m.n_ref
+        self._gpu = {
+            "torch": torch,
+            "ext": ext,
+            "dev": dev,
+            "w": up(m.weight),
+            "medians": up(m.medians),
+            "inv_scale": up(m.inv_scale),
+            # THE big buffer: [F, n_ref] sorted reference in HBM
+            "ref_sorted": up(m.ref_sorted).reshape(-1),
+            "rs_off": torch.from_numpy(rs_off).to(dev),
+        }
+
+    def hbm_bytes(self) -> int:
+        """Resident model + drift-reference footprint."""
+        m = self.model
+        return int(m.ref_sorted.size * 4 + m.weight.size * 12)
+
+    def score_arrays(self, x: np.ndarray, with_drift: bool = True) -> dict:
+        if self.device == "cuda":
+            return self._score_gpu(x, with_drift)
+        return self._score_cpu(x, with_drift)
+
+    # ------------------------------------------------------------------ GPU
+    def _score_gpu(self, x: np.ndarray, with_drift: bool) -> dict:
+        g = self._gpu
+        torch = g["torch"]
+        b = len(x)
+        xt = torch.from_numpy(np.ascontiguousarray(x, dtype=np.float32)).to(
+            g["dev"], non_blocking=True
+        )
+        proba, iscore, outlier = g["ext"].dense_score(
+            xt, g["medians"], g["inv_scale"], g["w"],
+            float(self.model.bias), float(self.model.z_threshold),
+        )
+        out = {}
+        if with_drift:
+            db = min(b, self.MAX_DRIFT_ROWS)
+            ks_d = g["ext"].ks_stats(
+                xt[:db].contiguous(), g["medians"], g["ref_sorted"], g["rs_off"]
+            )
+            out["ks_d"] = ks_d.cpu().numpy()
+        out.update(
+            predictions=proba.cpu().numpy(),
+            instance_score=iscore.cpu().numpy(),
+            outliers=outlier.cpu().numpy(),
+        )
+        if with_drift:
+            out["p_vals"] = self._pvals(out["ks_d"], min(b, self.MAX_DRIFT_ROWS))
+        return out
+
+    # ------------------------------------------------------------------ CPU
+    def _score_cpu(self, x: np.ndarray, with_drift: bool) -> dict:
+        m = self.model
+        x = np.asarray(x, dtype=np.float32)
+        xi = np.where(np.isnan(x), m.medians[None, :], x)
+        logit = xi.astype(np.float64) @ m.weight.astype(np.float64) + m.bias
+        proba = 1.0 / (1.0 + np.exp(-logit))
+        z = np.abs(xi - m.medians[None, :]) * m.inv_scale[None, :]
+        zmax = z.max(axis=1).astype(np.float64)
+        out = {
+            "predictions": proba,
+            "instance_score": zmax,
+            "outliers": (zmax > m.z_threshold).astype(np.float64),
+        }
+        if with_drift:
+            from .models.drift import ks_2samp_d
+
+            db = min(len(x), self.MAX_DRIFT_ROWS)
+            ks_d = np.array(
+                [
+                    ks_2samp_d(m.ref_sorted[j].astype(np.float64), xi[:db, j])
+                    for j in range(m.n_features)
+                ],
+                dtype=np.float32,
+            )
+            out["ks_d"] = ks_d
+            out["p_vals"] = self._pvals(ks_d, db)
+        return out
+
+    def _pvals(self, ks_d: np.ndarray, n_batch: int) -> np.ndarray:
+        from .models.drift import ks_asymp_pvalue_many
+
+        return ks_asymp_pvalue_many(ks_d.astype(np.float64), self.model.n_ref, n_batch)
