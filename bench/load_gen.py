@@ -19,6 +19,11 @@ import time
 
 import numpy as np
 
+# runnable from the repo root or anywhere: put the repo on sys.path
+import os as _os
+import sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 
 async def worker(client, url, body, stop_at, stats):
     while time.perf_counter() < stop_at:
