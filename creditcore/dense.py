@@ -235,8 +235,12 @@ class DenseEngine:
 
     def hbm_bytes(self) -> int:
         """Resident model + drift-reference footprint."""
+
+        def numel(a):
+            return a.numel() if hasattr(a, "numel") else a.size
+
         m = self.model
-        return int(m.ref_sorted.size * 4 + m.weight.size * 12)
+        return int(numel(m.ref_sorted) * 4 + numel(m.weight) * 12)
 
     def score_arrays(self, x: np.ndarray, with_drift: bool = True) -> dict:
         if self.device == "cuda":
