@@ -262,9 +262,16 @@ class DenseEngine:
         out = {}
         if with_drift:
             db = min(b, self.MAX_DRIFT_ROWS)
-            ks_d = g["ext"].ks_stats(
-                xt[:db].contiguous(), g["medians"], g["ref_sorted"], g["rs_off"]
-            )
+            if db >= 4096:
+                # large batches: rocPRIM column sort (torch.sort), then the
+                # chip-filling scan kernel (F x row-chunk blocks, atomicMax)
+                xi = torch.where(torch.isnan(xt[:db]), g["medians"][None, :], xt[:db])
+                xs = xi.sort(dim=0).values.contiguous()
+                ks_d = g["ext"].ks_stats_sorted(xs, g["ref_sorted"], g["rs_off"])
+            else:
+                ks_d = g["ext"].ks_stats(
+                    xt[:db].contiguous(), g["medians"], g["ref_sorted"], g["rs_off"]
+                )
             out["ks_d"] = ks_d.cpu().numpy()
         out.update(
             predictions=proba.cpu().numpy(),

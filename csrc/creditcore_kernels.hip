@@ -461,6 +461,69 @@ std::vector<torch::Tensor> dense_score(
   return {proba, iscore, outlier};
 }
 
+// Scan-only K-S for pre-sorted batch columns (large-B dense path: the sort
+// goes through rocPRIM (torch.sort), then this kernel fills the chip with
+// (feature x row-chunk) blocks; per-feature max via f32-bit atomicMax).
+__global__ __launch_bounds__(BLOCK) void ks_scan_kernel(
+    const float* __restrict__ xs,         // [B, F] column-sorted
+    int n_cols,
+    int n_rows,
+    const float* __restrict__ ref_sorted,
+    const int* __restrict__ rs_off,
+    unsigned int* __restrict__ ks_bits)   // [F] f32 bits, pre-zeroed
+{
+  const int j = blockIdx.x;
+  const int i = blockIdx.y * blockDim.x + threadIdx.x;
+  const int m = n_rows;
+  const int lo = rs_off[j];
+  const int n = rs_off[j + 1] - lo;
+  const float* __restrict__ ref = ref_sorted + lo;
+
+  double dmax = 0.0;
+  if (i < m) {
+    const float b = xs[(size_t)i * n_cols + j];
+    int l = 0, r = n;
+    while (l < r) { const int mid = (l + r) >> 1; if (ref[mid] < b) l = mid + 1; else r = mid; }
+    const int sl = l;
+    r = n;
+    while (l < r) { const int mid = (l + r) >> 1; if (ref[mid] <= b) l = mid + 1; else r = mid; }
+    const int sr = l;
+    int bl = 0; r = i;  // tie-run bounds within the sorted column
+    while (bl < r) { const int mid = (bl + r) >> 1;
+      if (xs[(size_t)mid * n_cols + j] < b) bl = mid + 1; else r = mid; }
+    int br = i + 1; r = m;
+    while (br < r) { const int mid = (br + r) >> 1;
+      if (xs[(size_t)mid * n_cols + j] <= b) br = mid + 1; else r = mid; }
+    const double fl = fabs((double)sl / n - (double)bl / m);
+    const double fr = fabs((double)sr / n - (double)br / m);
+    dmax = fmax(fl, fr);
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    dmax = fmax(dmax, __shfl_down(dmax, off, 64));
+  if ((threadIdx.x & 63) == 0 && dmax > 0.0)
+    atomicMax(&ks_bits[j], __float_as_uint((float)dmax));
+}
+
+torch::Tensor ks_stats_sorted(
+    torch::Tensor xs_sorted, torch::Tensor ref_sorted, torch::Tensor rs_off)
+{
+  TORCH_CHECK(xs_sorted.is_cuda() && xs_sorted.scalar_type() == torch::kFloat32
+              && xs_sorted.is_contiguous());
+  const int B = (int)xs_sorted.size(0);
+  const int F = (int)xs_sorted.size(1);
+  TORCH_CHECK((int)rs_off.size(0) == F + 1, "rs_off size mismatch");
+  auto bits = torch::zeros({F},
+      torch::TensorOptions().dtype(torch::kInt32).device(xs_sorted.device()));
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(ks_scan_kernel, dim3(F, ceil_div(B, BLOCK)), dim3(BLOCK), 0, stream,
+      xs_sorted.data_ptr<float>(), F, B,
+      ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
+      reinterpret_cast<unsigned int*>(bits.data_ptr<int>()));
+  HIP_CHECK(hipGetLastError());
+  return bits.view(torch::kFloat32);
+}
+
 // Generic exact K-S D over any column count (dense drift path; the credit
 // path embeds the same kernel in its session graph).
 torch::Tensor ks_stats(
@@ -1247,6 +1310,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       });
   m.def("dense_score", &dense_score,
         "Fused impute + logistic linear score + robust-z outlier (gfx950)");
+  m.def("ks_stats_sorted", &ks_stats_sorted,
+        "Exact K-S D per column for pre-sorted batch columns (gfx950)");
   m.def("ks_stats", &ks_stats,
         "Exact two-sample K-S D per column, any column count (gfx950)");
   m.def("drift_pvals_host", &drift_pvals_host,
