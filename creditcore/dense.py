@@ -263,11 +263,15 @@ class DenseEngine:
         if with_drift:
             db = min(b, self.MAX_DRIFT_ROWS)
             if db >= 4096:
-                # large batches: rocPRIM column sort (torch.sort), then the
-                # chip-filling scan kernel (F x row-chunk blocks, atomicMax)
-                xi = torch.where(torch.isnan(xt[:db]), g["medians"][None, :], xt[:db])
-                xs = xi.sort(dim=0).values.contiguous()
-                ks_d = g["ext"].ks_stats_sorted(xs, g["ref_sorted"], g["rs_off"])
+                # large batches: impute+transpose to [F, B] (contiguous
+                # per-feature rows), rocPRIM segmented sort along rows, then
+                # the chip-filling scan kernel (F x row-chunk blocks,
+                # atomicMax) with fully coalesced accesses
+                xi_t = torch.where(
+                    torch.isnan(xt[:db].t()), g["medians"][:, None], xt[:db].t()
+                )
+                xs = xi_t.sort(dim=1).values
+                ks_d = g["ext"].ks_stats_sorted(xs.contiguous(), g["ref_sorted"], g["rs_off"])
             else:
                 ks_d = g["ext"].ks_stats(
                     xt[:db].contiguous(), g["medians"], g["ref_sorted"], g["rs_off"]

@@ -465,7 +465,7 @@ std::vector<torch::Tensor> dense_score(
 // goes through rocPRIM (torch.sort), then this kernel fills the chip with
 // (feature x row-chunk) blocks; per-feature max via f32-bit atomicMax).
 __global__ __launch_bounds__(BLOCK) void ks_scan_kernel(
-    const float* __restrict__ xs,         // [B, F] column-sorted
+    const float* __restrict__ xs,         // [F, B] row-contiguous, each row sorted
     int n_cols,
     int n_rows,
     const float* __restrict__ ref_sorted,
@@ -478,10 +478,11 @@ __global__ __launch_bounds__(BLOCK) void ks_scan_kernel(
   const int lo = rs_off[j];
   const int n = rs_off[j + 1] - lo;
   const float* __restrict__ ref = ref_sorted + lo;
+  const float* __restrict__ col = xs + (size_t)j * n_rows;  // contiguous
 
   double dmax = 0.0;
   if (i < m) {
-    const float b = xs[(size_t)i * n_cols + j];
+    const float b = col[i];
     int l = 0, r = n;
     while (l < r) { const int mid = (l + r) >> 1; if (ref[mid] < b) l = mid + 1; else r = mid; }
     const int sl = l;
@@ -490,10 +491,10 @@ __global__ __launch_bounds__(BLOCK) void ks_scan_kernel(
     const int sr = l;
     int bl = 0; r = i;  // tie-run bounds within the sorted column
     while (bl < r) { const int mid = (bl + r) >> 1;
-      if (xs[(size_t)mid * n_cols + j] < b) bl = mid + 1; else r = mid; }
+      if (col[mid] < b) bl = mid + 1; else r = mid; }
     int br = i + 1; r = m;
     while (br < r) { const int mid = (br + r) >> 1;
-      if (xs[(size_t)mid * n_cols + j] <= b) br = mid + 1; else r = mid; }
+      if (col[mid] <= b) br = mid + 1; else r = mid; }
     const double fl = fabs((double)sl / n - (double)bl / m);
     const double fr = fabs((double)sr / n - (double)br / m);
     dmax = fmax(fl, fr);
@@ -510,8 +511,8 @@ torch::Tensor ks_stats_sorted(
 {
   TORCH_CHECK(xs_sorted.is_cuda() && xs_sorted.scalar_type() == torch::kFloat32
               && xs_sorted.is_contiguous());
-  const int B = (int)xs_sorted.size(0);
-  const int F = (int)xs_sorted.size(1);
+  const int F = (int)xs_sorted.size(0);  // [F, B]: per-feature sorted rows
+  const int B = (int)xs_sorted.size(1);
   TORCH_CHECK((int)rs_off.size(0) == F + 1, "rs_off size mismatch");
   auto bits = torch::zeros({F},
       torch::TensorOptions().dtype(torch::kInt32).device(xs_sorted.device()));
