@@ -193,7 +193,7 @@ class DenseEngine:
     identical semantics (golden tests compare the two).
     """
 
-    MAX_DRIFT_ROWS = 16384
+    MAX_DRIFT_ROWS = 65536  # sort+scan path; the in-LDS kernel caps at 16384
 
     def __init__(self, model: DenseModel, device: str = "cpu", device_index: int = 0):
         self.model = model
@@ -262,7 +262,7 @@ class DenseEngine:
         out = {}
         if with_drift:
             db = min(b, self.MAX_DRIFT_ROWS)
-            if db >= 4096:
+            if db > 16384:
                 # large batches: impute+transpose to [F, B] (contiguous
                 # per-feature rows), rocPRIM segmented sort along rows, then
                 # the chip-filling scan kernel (F x row-chunk blocks,

@@ -87,6 +87,25 @@ def test_dense_gpu_parity(dense_model):
 
 
 @pytest.mark.gpu
+def test_dense_gpu_scan_ks_parity():
+    """Large-batch drift path (B > 16384: rocPRIM sort + ks_scan_kernel)
+    vs the CPU exact reference."""
+    model = train_dense(
+        n_rows=50_000, n_feats=32, ref_rows=30_000, epochs=1,
+        batch_rows=8192, device="cpu", seed=11, log=lambda *a: None,
+    )
+    eng_g = DenseEngine(model, device="cuda")
+    eng_c = DenseEngine(model, device="cpu")
+    rng = np.random.default_rng(13)
+    x = rng.standard_normal((20_000, 32)).astype(np.float32)
+    x[:50] = x[:1]  # tie runs exercise the tie-bound searches
+    g = eng_g.score_arrays(x)
+    c = eng_c.score_arrays(x)
+    np.testing.assert_allclose(g["ks_d"], c["ks_d"], atol=1e-6)
+    np.testing.assert_allclose(g["p_vals"], c["p_vals"], atol=1e-5)
+
+
+@pytest.mark.gpu
 def test_dense_gpu_train_and_score():
     """On-GPU training (keep_on_device): reference stays in HBM."""
     model = train_dense(
