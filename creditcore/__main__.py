@@ -25,9 +25,18 @@ def _cmd_train(argv):
     p.add_argument("--n-rows", type=int, default=20_000)
     p.add_argument("--seed", type=int, default=2024)
     p.add_argument("--no-register", action="store_true")
+    p.add_argument("--data", default=None,
+                   help="CSV with the UCI schema (the reference's curated "
+                        "table, 00-create-external-table.ipynb); synthetic "
+                        "data is generated when omitted")
     a = p.parse_args(argv)
     from .train import train_and_register
 
+    df = None
+    if a.data:
+        import pandas as pd
+
+        df = pd.read_csv(a.data)
     uri = train_and_register(
         model_dir=a.model_dir,
         model_name=a.model_name,
@@ -36,6 +45,7 @@ def _cmd_train(argv):
         n_rows=a.n_rows,
         seed=a.seed,
         register=not a.no_register,
+        df=df,
     )
     print(uri)
 
