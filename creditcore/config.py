@@ -31,6 +31,7 @@ class ServeConfig:
     service_name: str = field(default_factory=lambda: _env("service_name", "credit-default-api"))
     host: str = field(default_factory=lambda: _env("host", "0.0.0.0"))
     port: int = field(default_factory=lambda: _env("port", 5000, int))
+    workers: int = field(default_factory=lambda: _env("workers", 1, int))
 
     # device
     device: str = field(default_factory=lambda: _env("device", "auto"))  # auto|cuda|cpu

@@ -152,11 +152,16 @@ class DriftSync:
     def accumulate(self, cat_hist, nums) -> None:
         """Fold one scored batch into the local histogram.
 
-        ``cat_hist``: int tensor [C] (the drift kernel's output);
-        ``nums``: float32 tensor [B, 14] (raw, NaNs allowed).
+        ``cat_hist``: int array/tensor [C] (the drift kernel's output);
+        ``nums``: float32 array/tensor [B, 14] (raw, NaNs allowed).
+        numpy inputs take a numpy fast path (per-request torch-CPU op
+        overhead was ~6 ms at bs=1024); tensors keep the torch path (GPU).
         """
         import torch
 
+        if not torch.is_tensor(nums):
+            self._accumulate_np(np.asarray(cat_hist), np.asarray(nums))
+            return
         self.local[: self.C] += cat_hist.to(self.local.device, torch.int64)
         x = nums.to(self.edges.device).t().contiguous()  # [N_NUM, B]
         x = torch.where(torch.isnan(x), self.medians_t[:, None], x)
@@ -165,6 +170,21 @@ class DriftSync:
         flat = (idx + offs).reshape(-1)
         binc = torch.bincount(flat, minlength=N_NUM * self.n_bins)
         self.local[self.C :] += binc.to(self.local.device)
+        self.batches += 1
+
+    def _accumulate_np(self, cat_hist: np.ndarray, nums: np.ndarray) -> None:
+        if getattr(self, "_edges_np", None) is None:
+            self._edges_np = self.edges.cpu().numpy()
+            self._medians_np = self.medians_t.cpu().numpy()
+        local = self.local.numpy()
+        local[: self.C] += cat_hist.astype(np.int64)
+        x = np.where(np.isnan(nums), self._medians_np[None, :], nums)
+        offs = np.arange(N_NUM) * self.n_bins
+        idx = np.empty((N_NUM, len(x)), dtype=np.int64)
+        for j in range(N_NUM):
+            idx[j] = np.searchsorted(self._edges_np[j], x[:, j], side="right")
+        binc = np.bincount((idx + offs[:, None]).reshape(-1), minlength=N_NUM * self.n_bins)
+        local[self.C :] += binc
         self.batches += 1
 
     def allreduce(self, group=None) -> None:

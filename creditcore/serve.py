@@ -88,8 +88,6 @@ class ReplicaPool:
 
 @asynccontextmanager
 async def lifespan(app: FastAPI):
-    import torch
-
     from .parallel import DriftSync
 
     cfg: ServeConfig = app.state.cfg
@@ -102,10 +100,7 @@ async def lifespan(app: FastAPI):
         def run(codes, nums):
             out = e.score_arrays(codes, nums)
             if "cat_hist" in out:
-                drift_sync.accumulate(
-                    torch.from_numpy(np.ascontiguousarray(out["cat_hist"])),
-                    torch.from_numpy(nums),
-                )
+                drift_sync.accumulate(out["cat_hist"], nums)
                 if drift_sync.batches % max(cfg.drift_sync_period, 1) == 0:
                     drift_sync.allreduce()
                     state["metrics"].observe_drift_sync()
@@ -271,6 +266,11 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
     return app
 
 
+def app_from_env() -> FastAPI:
+    """uvicorn factory target for multi-worker serving (config from env)."""
+    return create_app(ServeConfig())
+
+
 def main(argv: list[str] | None = None):
     import logging
 
@@ -279,7 +279,25 @@ def main(argv: list[str] | None = None):
     logging.basicConfig(level=logging.INFO)
     cfg = ServeConfig.from_args(argv)
     app = create_app(cfg)
-    uvicorn.run(app, host=cfg.host, port=cfg.port, log_level="info")
+    # access_log off: the app writes its own two JSON lines per request
+    if cfg.workers > 1:
+        # process-level scaling: each worker owns its engines and GIL
+        # (SO_REUSEPORT fan-in by uvicorn). Config rides the env.
+        import dataclasses
+
+        for f in dataclasses.fields(cfg):
+            os.environ[f"CREDITCORE_{f.name.upper()}"] = str(getattr(cfg, f.name))
+        uvicorn.run(
+            "creditcore.serve:app_from_env",
+            factory=True,
+            host=cfg.host,
+            port=cfg.port,
+            log_level="info",
+            access_log=False,
+            workers=cfg.workers,
+        )
+    else:
+        uvicorn.run(app, host=cfg.host, port=cfg.port, log_level="info", access_log=False)
 
 
 if __name__ == "__main__":

@@ -19,15 +19,14 @@ logger = logging.getLogger("creditcore.requests")
 
 
 def log_inference_data(service_name: str, request_id: str, data_json: str) -> None:
+    """``data_json`` must already be JSON (the raw request body): it is
+    embedded verbatim as a nested JSON value instead of being re-escaped
+    into a string — re-escaping a ~0.6 MB body cost ~4 ms per request."""
     logger.info(
-        json.dumps(
-            {
-                "service_name": service_name,
-                "type": "InferenceData",
-                "request_id": request_id,
-                "data": data_json,
-            }
-        )
+        '{"service_name": %s, "type": "InferenceData", "request_id": "%s", "data": %s}',
+        json.dumps(service_name),
+        request_id,
+        data_json,
     )
 
 
