@@ -73,13 +73,52 @@ async def run(args):
     print(json.dumps(out))
 
 
+def _proc_worker(args, q):
+    import io
+    import contextlib
+
+    buf = io.StringIO()
+    with contextlib.redirect_stdout(buf):
+        asyncio.run(run(args))
+    q.put(buf.getvalue())
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--url", default="http://127.0.0.1:5000")
     p.add_argument("--rows", type=int, default=1024)
     p.add_argument("--concurrency", type=int, default=8)
     p.add_argument("--duration", type=float, default=20.0)
-    asyncio.run(run(p.parse_args()))
+    p.add_argument("--processes", type=int, default=1,
+                   help="client processes (a single event loop saturates "
+                        "before the server does on large bodies)")
+    args = p.parse_args()
+    if args.processes <= 1:
+        asyncio.run(run(args))
+        return
+    import copy
+    import multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    sub = copy.copy(args)
+    sub.concurrency = max(1, args.concurrency // args.processes)
+    procs = [ctx.Process(target=_proc_worker, args=(sub, q)) for _ in range(args.processes)]
+    for pr in procs:
+        pr.start()
+    results = [json.loads(q.get(timeout=args.duration + 120)) for _ in procs]
+    for pr in procs:
+        pr.join(timeout=30)
+    total = sum(r["value"] for r in results)
+    out = dict(results[0])
+    out["value"] = round(total, 2)
+    out["rows_per_sec"] = round(total * args.rows, 1)
+    out["concurrency"] = sub.concurrency * args.processes
+    out["client_processes"] = args.processes
+    for k in ("latency_ms_p50", "latency_ms_p90", "latency_ms_p99"):
+        vals = [r[k] for r in results if r[k] is not None]
+        out[k] = round(float(np.median(vals)), 2) if vals else None
+    print(json.dumps(out))
 
 
 if __name__ == "__main__":

@@ -36,6 +36,7 @@
 #define N_CAT 9
 #define N_NUM 14
 #define MAX_DRIFT_ROWS 16384  // LDS cap for the K-S sort (64 KiB of f32)
+static constexpr size_t KS_LDS_BYTES_MAX = 160 * 1024 - 1024;  // gfx950 LDS/CU
 
 #define HIP_CHECK(expr)                                              \
   do {                                                               \
@@ -529,22 +530,32 @@ torch::Tensor ks_stats_sorted(
 // path embeds the same kernel in its session graph).
 torch::Tensor ks_stats(
     torch::Tensor nums, torch::Tensor medians,
-    torch::Tensor ref_sorted, torch::Tensor rs_off)
+    torch::Tensor ref_sorted, torch::Tensor rs_off,
+    int64_t block, int64_t ref_lds)
 {
   TORCH_CHECK(nums.is_cuda() && nums.scalar_type() == torch::kFloat32 && nums.is_contiguous());
   const int B = (int)nums.size(0);
   const int F = (int)nums.size(1);
   TORCH_CHECK(B <= MAX_DRIFT_ROWS, "K-S batch too large: ", B);
   TORCH_CHECK((int)rs_off.size(0) == F + 1, "rs_off size mismatch");
+  TORCH_CHECK(block == 256 || block == 512 || block == 1024, "bad block");
   auto ks_d = torch::empty({F},
       torch::TensorOptions().dtype(torch::kFloat32).device(nums.device()));
   hipStream_t stream = c10::hip::getCurrentHIPStream();
   int m_pow2 = 2;
   while (m_pow2 < B) m_pow2 <<= 1;
-  hipLaunchKernelGGL(ks_kernel, dim3(F), dim3(BLOCK),
-      (size_t)m_pow2 * sizeof(float), stream,
+  size_t smem = (size_t)m_pow2 * sizeof(float);
+  if (ref_lds) {
+    int64_t maxr = 0;
+    auto ro = rs_off.cpu();
+    auto* rp = ro.data_ptr<int>();
+    for (int j = 0; j < F; ++j) maxr = std::max<int64_t>(maxr, rp[j + 1] - rp[j]);
+    smem += (size_t)maxr * sizeof(float);
+    TORCH_CHECK(smem <= KS_LDS_BYTES_MAX, "ref too large for LDS");
+  }
+  hipLaunchKernelGGL(ks_kernel, dim3(F), dim3((int)block), smem, stream,
       nums.data_ptr<float>(), medians.data_ptr<float>(), F, B, m_pow2,
-      /*ref_lds=*/0, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
+      (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
       ks_d.data_ptr<float>());
   HIP_CHECK(hipGetLastError());
   return ks_d;
@@ -692,8 +703,11 @@ struct ScoreSession {
       // CU budget; searches then stay on-chip.
       const size_t batch_bytes = (size_t)m_pow2 * sizeof(float);
       const size_t ref_bytes = (size_t)max_ref_len * sizeof(float);
-      const bool ref_lds = batch_bytes + ref_bytes <= KS_LDS_BYTES;
-      hipLaunchKernelGGL(ks_kernel, dim3(N_NUM), dim3(BLOCK),
+      // staging the ref column only pays when the batch searches it enough
+      const bool ref_lds =
+          b >= 4096 && batch_bytes + ref_bytes <= KS_LDS_BYTES;
+      const int ks_block = (m_pow2 >= 2048) ? 512 : BLOCK;
+      hipLaunchKernelGGL(ks_kernel, dim3(N_NUM), dim3(ks_block),
           batch_bytes + (ref_lds ? ref_bytes : 0), stream2,
           d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b, m_pow2,
           (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
@@ -744,7 +758,7 @@ struct ScoreSession {
 
   std::unordered_map<uint64_t, hipGraphExec_t> graphs;
   int64_t max_ref_len{};
-  static constexpr size_t KS_LDS_BYTES = 160 * 1024 - 1024;
+  static constexpr size_t KS_LDS_BYTES = KS_LDS_BYTES_MAX;
 
   void synchronize() {
     py::gil_scoped_release nogil;
@@ -1314,7 +1328,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ks_stats_sorted", &ks_stats_sorted,
         "Exact K-S D per column for pre-sorted batch columns (gfx950)");
   m.def("ks_stats", &ks_stats,
-        "Exact two-sample K-S D per column, any column count (gfx950)");
+        "Exact two-sample K-S D per column, any column count (gfx950)",
+        py::arg("nums"), py::arg("medians"), py::arg("ref_sorted"),
+        py::arg("rs_off"), py::arg("block") = 256, py::arg("ref_lds") = 0);
   m.def("drift_pvals_host", &drift_pvals_host,
         "Drift p-values from kernel statistics (chi2 + Pelz-Good K-S), host C");
   py::class_<ScoreSession>(m, "ScoreSession")
