@@ -130,18 +130,25 @@ def main():
         # released in the C parser) overlaps request i's GPU work. Every
         # request is still fully processed — parse, encode, kernels, drift
         # p-values, response assembly.
+        from collections import deque
         from concurrent.futures import ThreadPoolExecutor
 
-        executor = ThreadPoolExecutor(max_workers=1)
+        executor = ThreadPoolExecutor(max_workers=2)
+        DEPTH = 3  # encode i+1..i+3 overlap scoring of i (C parser drops the GIL)
 
         def run_steps(k: int):
             outs = []
-            fut = executor.submit(engine.encode_json_body, pool[0])
+            q = deque(
+                executor.submit(engine.encode_json_body, pool[i % len(pool)])
+                for i in range(min(DEPTH, k))
+            )
             for i in range(k):
-                codes, nums = fut.result()
-                fut = executor.submit(
-                    engine.encode_json_body, pool[(i + 1) % len(pool)]
-                )
+                codes, nums = q.popleft().result()
+                nxt = i + DEPTH
+                if nxt < k:
+                    q.append(
+                        executor.submit(engine.encode_json_body, pool[nxt % len(pool)])
+                    )
                 out = engine._score_encoded(codes, nums)
                 assert out["rows"] == args.rows
                 outs.append(out)

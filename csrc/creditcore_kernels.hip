@@ -538,7 +538,7 @@ torch::Tensor ks_stats(
   const int F = (int)nums.size(1);
   TORCH_CHECK(B <= MAX_DRIFT_ROWS, "K-S batch too large: ", B);
   TORCH_CHECK((int)rs_off.size(0) == F + 1, "rs_off size mismatch");
-  TORCH_CHECK(block == 256 || block == 512 || block == 1024, "bad block");
+  TORCH_CHECK(block == 256, "ks_kernel is __launch_bounds__(256)");
   auto ks_d = torch::empty({F},
       torch::TensorOptions().dtype(torch::kFloat32).device(nums.device()));
   hipStream_t stream = c10::hip::getCurrentHIPStream();
@@ -703,11 +703,12 @@ struct ScoreSession {
       // CU budget; searches then stay on-chip.
       const size_t batch_bytes = (size_t)m_pow2 * sizeof(float);
       const size_t ref_bytes = (size_t)max_ref_len * sizeof(float);
-      // staging the ref column only pays when the batch searches it enough
-      const bool ref_lds =
-          b >= 4096 && batch_bytes + ref_bytes <= KS_LDS_BYTES;
-      const int ks_block = (m_pow2 >= 2048) ? 512 : BLOCK;
-      hipLaunchKernelGGL(ks_kernel, dim3(N_NUM), dim3(ks_block),
+      // measured (bench/kernel_micro.py on MI355X): staging the ref column
+      // in LDS is slower at b=1024 (62 vs 48 us) and within noise at 16k;
+      // ks_kernel is __launch_bounds__(256), so block stays 256.
+      (void)ref_bytes;
+      const bool ref_lds = false;
+      hipLaunchKernelGGL(ks_kernel, dim3(N_NUM), dim3(BLOCK),
           batch_bytes + (ref_lds ? ref_bytes : 0), stream2,
           d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b, m_pow2,
           (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
