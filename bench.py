@@ -149,7 +149,7 @@ def main():
                     q.append(
                         executor.submit(engine.encode_json_body, pool[nxt % len(pool)])
                     )
-                out = engine._score_encoded(codes, nums)
+                out = engine.score_encoded_bytes(codes, nums)
                 assert out["rows"] == args.rows
                 outs.append(out)
             return outs

@@ -183,6 +183,40 @@ class ScoringEngine:
             )
         return self._default_rows
 
+    def score_json_bytes(self, body: bytes) -> dict:
+        """Wire-format in, wire-format out: parse the JSON request with the
+        C parser, score through the session graph, and serialize the
+        response JSON in C — the full /score answer with no Python object
+        pass. Returns {"response_bytes", "rows"}."""
+        codes, nums = self.encode_json_body(body)
+        return self.score_encoded_bytes(codes, nums)
+
+    def score_encoded_bytes(self, codes: np.ndarray, nums: np.ndarray) -> dict:
+        b = len(codes)
+        if self.device != "cuda" or b == 0:
+            import json
+
+            out = self._score_encoded(codes, nums)
+            return {
+                "response_bytes": json.dumps(out["response"]).encode(),
+                "rows": b,
+            }
+        g = self._gpu
+        self._ensure_capacity(b)
+        drift_now = b <= self.DRIFT_MAX_ROWS
+        g["np_codes"][:b] = codes
+        g["np_nums"][:b] = nums
+        g["sess"].score(b, drift_now, True)
+        nb = b
+        if not drift_now:
+            g["sess"].score(self.DRIFT_MAX_ROWS, True, True)
+            nb = self.DRIFT_MAX_ROWS
+        pvals = cpu_ref.pvals_from_stats(self.packed, g["np_hist"], g["np_ksd"], nb)
+        resp = g["ext"].build_response_json(
+            g["sess"].pin_outs, b, np.ascontiguousarray(pvals), FEATURES
+        )
+        return {"response_bytes": resp, "rows": b}
+
     def score_records(self, records) -> dict:
         """Score a request body (list of dicts / DataFrame); returns the
         reference response shape (02-register cell-9)."""

@@ -1309,6 +1309,65 @@ py::tuple encode_json(py::bytes body, py::list vocabs, py::list cat_names,
   return encode_json_impl(st, body);
 }
 
+// ---------------------------------------------------------------------------
+// Response serializer: the reference-shaped /score response straight to
+// JSON bytes (shortest-round-trip doubles via std::to_chars). Skips the
+// Python dict + json.dumps pass on the serving hot path.
+// ---------------------------------------------------------------------------
+
+static inline void append_double(std::string& out, double v) {
+  char buf[32];
+  auto [p, ec] = std::to_chars(buf, buf + sizeof(buf), v);
+  (void)ec;
+  // JSON requires a fraction or exponent for floats; Python emits "0.0"
+  // style. Keep parity: append ".0" to bare integers.
+  bool has_dot = false;
+  for (char* c = buf; c < p; ++c)
+    if (*c == '.' || *c == 'e' || *c == 'E' || *c == 'n' || *c == 'i') { has_dot = true; break; }
+  out.append(buf, p);
+  if (!has_dot) out += ".0";
+}
+
+py::bytes build_response_json(py::object pin_outs, int64_t b,
+                              py::array_t<double> pvals, py::list feature_names) {
+  auto outs = py::cast<torch::Tensor>(pin_outs);
+  const double* base = outs.data_ptr<double>();
+  const double* proba = base;
+  const double* outlier = base + 2 * b;
+  const int nf = (int)py::len(feature_names);
+  const double* pv = pvals.data();
+
+  std::vector<std::string> names(nf);
+  for (int j = 0; j < nf; ++j) names[j] = py::cast<std::string>(feature_names[j]);
+
+  std::string out;
+  {
+    py::gil_scoped_release nogil;
+    out.reserve((size_t)b * 24 + 2048);
+    out += "{\"predictions\": [";
+    for (int64_t i = 0; i < b; ++i) {
+      if (i) out += ", ";
+      append_double(out, proba[i]);
+    }
+    out += "], \"outliers\": [";
+    for (int64_t i = 0; i < b; ++i) {
+      if (i) out += ", ";
+      out += (outlier[i] != 0.0) ? "1.0" : "0.0";
+    }
+    out += "], \"feature_drift_batch\": {";
+    for (int j = 0; j < nf; ++j) {
+      if (j) out += ", ";
+      out += '\"';
+      out += names[j];
+      out += "\": ";
+      // (1 - p) computed in float32, reference 02-register cell-9 semantics
+      append_double(out, (double)(1.0f - (float)pv[j]));
+    }
+    out += "}}";
+  }
+  return py::bytes(out);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("score_forest_pipeline", &score_forest_pipeline,
         "Forest classifier + isolation forest scoring (gfx950)");
@@ -1332,6 +1391,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Exact two-sample K-S D per column, any column count (gfx950)",
         py::arg("nums"), py::arg("medians"), py::arg("ref_sorted"),
         py::arg("rs_off"), py::arg("block") = 256, py::arg("ref_lds") = 0);
+  m.def("build_response_json", &build_response_json,
+        "Serialize the /score response to JSON bytes (C, shortest doubles)");
   m.def("drift_pvals_host", &drift_pvals_host,
         "Drift p-values from kernel statistics (chi2 + Pelz-Good K-S), host C");
   py::class_<ScoreSession>(m, "ScoreSession")

@@ -102,6 +102,42 @@ def test_empty_array(ext, packed, defaults):
     assert c.shape == (0, 9) and n.shape == (0, 14)
 
 
+def test_response_bytes_matches_dict_path(packed):
+    """score_json_bytes (C serializer) must produce the same JSON values as
+    the dict path at double precision."""
+    import json
+
+    from creditcore.data import make_request_batch
+    from creditcore.engine import ScoringEngine
+
+    eng = ScoringEngine(packed, device="cpu")
+    recs = make_request_batch(64, seed=21)
+    body = json.dumps(recs).encode()
+    d = eng.score_json(body)["response"]
+    # CPU path serializes via json.dumps; on GPU the C path is exercised by
+    # the gpu-marked test below. Round-trip equality check:
+    b = json.loads(eng.score_json_bytes(body)["response_bytes"])
+    assert b == json.loads(json.dumps(d))
+
+
+@pytest.mark.gpu
+def test_response_bytes_gpu_matches_dict_path(packed):
+    import json
+
+    from creditcore.data import make_request_batch
+    from creditcore.engine import ScoringEngine
+
+    eng = ScoringEngine(packed, device="cuda")
+    recs = make_request_batch(257, seed=22)
+    body = json.dumps(recs).encode()
+    d = eng.score_json(body)["response"]
+    b = json.loads(eng.score_json_bytes(body)["response_bytes"])
+    np.testing.assert_allclose(b["predictions"], d["predictions"], rtol=0, atol=0)
+    np.testing.assert_array_equal(b["outliers"], d["outliers"])
+    for k, v in d["feature_drift_batch"].items():
+        assert abs(b["feature_drift_batch"][k] - v) < 1e-12
+
+
 def test_serving_fallback_handles_lax_types(model_dir):
     """A numeric string coerces through the pydantic fallback exactly as the
     reference would (fast path rejects, fallback accepts)."""
