@@ -54,14 +54,13 @@ def main():
         recs = make_request_batch(b, seed=b)
         codes, nums = encode_batch(recs, p.vocabs)
         d_nums = torch.from_numpy(nums).to(dev)
-        for block in (256, 512, 1024):
-            for ref_lds in (0, 1):
-                key = f"ks b={b} block={block} ref_lds={ref_lds}"
-                try:
-                    us = timed(lambda: ext.ks_stats(d_nums, medians, ref, rs_off, block, ref_lds))
-                    results[key] = round(us, 2)
-                except Exception as exc:
-                    results[key] = f"error: {exc}"
+        for block in (256, 512):
+            key = f"ks b={b} block={block}"
+            try:
+                us = timed(lambda: ext.ks_stats(d_nums, medians, ref, rs_off, block, 0))
+                results[key] = round(us, 2)
+            except Exception as exc:
+                results[key] = f"error: {exc}"
 
     # forest via the functional pipeline (classifier + iforest + finalize)
     cls_nodes = torch.from_numpy(np.ascontiguousarray(p.cls_nodes)).to(dev)
@@ -83,6 +82,17 @@ def main():
             )
         )
         results[f"forest_pipeline b={b}"] = round(us, 2)
+        # A/B: 1 vs 2 trees per thread (classifier forest alone)
+        for ilp in (0, 1):
+            us = timed(
+                lambda: ext.forest_ilp_bench(
+                    d_codes, d_nums, cls_nodes, cls_off, fc, fk, medians, ilp
+                )
+            )
+            results[f"forest_cls b={b} ilp={ilp}"] = round(us, 2)
+        a = ext.forest_ilp_bench(d_codes, d_nums, cls_nodes, cls_off, fc, fk, medians, 0)
+        c = ext.forest_ilp_bench(d_codes, d_nums, cls_nodes, cls_off, fc, fk, medians, 1)
+        results[f"forest_ilp_parity b={b}"] = float((a - c).abs().max().item())
 
     print(json.dumps(results, indent=2))
 

@@ -108,6 +108,82 @@ __global__ __launch_bounds__(BLOCK) void forest_kernel(
   atomicAdd(&acc[row], local);
 }
 
+// ILP variant: each thread walks TWO trees concurrently — two independent
+// node-gather chains per lane double the memory-level parallelism of the
+// latency-bound traversal (A/B'd against forest_kernel via kernel_micro).
+template <bool DIRECT>
+__global__ __launch_bounds__(BLOCK) void forest_kernel_ilp(
+    const short* __restrict__ codes,
+    const float* __restrict__ nums,
+    const float* __restrict__ medians,
+    const int4* __restrict__ nodes,
+    const int* __restrict__ tree_off,
+    int n_trees,
+    const int* __restrict__ feat_col,
+    const int* __restrict__ feat_code,
+    int n_rows,
+    double* __restrict__ acc)
+{
+  __shared__ short s_codes[N_CAT * BLOCK];
+  __shared__ float s_nums[N_NUM * BLOCK];
+  const int tid = threadIdx.x;
+  const int row = blockIdx.x * BLOCK + tid;
+  if (row < n_rows) {
+    if (!DIRECT) {
+#pragma unroll
+      for (int c = 0; c < N_CAT; ++c) s_codes[c * BLOCK + tid] = codes[row * N_CAT + c];
+    }
+#pragma unroll
+    for (int c = 0; c < N_NUM; ++c) {
+      const float v = nums[row * N_NUM + c];
+      s_nums[c * BLOCK + tid] = isnan(v) ? medians[c] : v;
+    }
+  }
+  if (row >= n_rows) return;
+
+  auto value_of = [&](int f) -> float {
+    if (DIRECT) return s_nums[f * BLOCK + tid];
+    const int col = feat_col[f];
+    const int code = feat_code[f];
+    return (code >= 0) ? ((s_codes[col * BLOCK + tid] == (short)code) ? 1.0f : 0.0f)
+                       : s_nums[col * BLOCK + tid];
+  };
+
+  double local = 0.0;
+  for (int t = blockIdx.y; t < n_trees; t += 2 * gridDim.y) {
+    const int ta = t;
+    const int tb = t + gridDim.y;
+    const int basea = tree_off[ta];
+    int4 na = nodes[basea];
+    bool la = true;
+    int baseb = 0;
+    int4 nb;
+    bool lb = tb < n_trees;
+    if (lb) { baseb = tree_off[tb]; nb = nodes[baseb]; }
+    while (la || lb) {
+      if (la) {
+        if (na.x >= 0) {
+          const float v = value_of(na.x);
+          na = nodes[basea + ((v <= __int_as_float(na.y)) ? na.z : na.w)];
+        } else {
+          local += (double)__int_as_float(na.y);
+          la = false;
+        }
+      }
+      if (lb) {
+        if (nb.x >= 0) {
+          const float v = value_of(nb.x);
+          nb = nodes[baseb + ((v <= __int_as_float(nb.y)) ? nb.z : nb.w)];
+        } else {
+          local += (double)__int_as_float(nb.y);
+          lb = false;
+        }
+      }
+    }
+  }
+  atomicAdd(&acc[row], local);
+}
+
 __global__ __launch_bounds__(BLOCK) void finalize_kernel(
     const double* __restrict__ cls_acc,
     const double* __restrict__ if_acc,
@@ -174,7 +250,8 @@ __global__ __launch_bounds__(BLOCK) void cat_hist_kernel(
 // CPU reference bitwise-closely. When the reference column fits the
 // remaining LDS (160 KiB/CU on gfx950) it is staged there too, so the
 // per-element binary searches hit LDS instead of bouncing off L2.
-__global__ __launch_bounds__(BLOCK) void ks_kernel(
+template <int BS>
+__global__ __launch_bounds__(BS) void ks_kernel_t(
     const float* __restrict__ nums,       // [B, n_cols]
     const float* __restrict__ medians,    // [n_cols]
     int n_cols,                           // feature count == gridDim.x
@@ -351,6 +428,38 @@ std::vector<torch::Tensor> score_forest_pipeline(
   return {proba, iscore, outlier};
 }
 
+torch::Tensor forest_ilp_bench(
+    torch::Tensor codes, torch::Tensor nums,
+    torch::Tensor nodes, torch::Tensor off,
+    torch::Tensor feat_col, torch::Tensor feat_code,
+    torch::Tensor medians, int64_t use_ilp)
+{
+  check_inputs(codes, nums);
+  const int B = (int)codes.size(0);
+  const int T = (int)off.size(0) - 1;
+  auto acc = torch::zeros({B},
+      torch::TensorOptions().dtype(torch::kFloat64).device(codes.device()));
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  const int row_blocks = ceil_div(B, BLOCK);
+  int chunks = std::max(1, std::min(ceil_div(2048, row_blocks), T));
+  if (use_ilp) {
+    chunks = std::max(1, std::min(ceil_div(2048, row_blocks), (T + 1) / 2));
+    hipLaunchKernelGGL((forest_kernel_ilp<false>), dim3(row_blocks, chunks),
+        dim3(BLOCK), 0, stream,
+        codes.data_ptr<short>(), nums.data_ptr<float>(), medians.data_ptr<float>(),
+        reinterpret_cast<const int4*>(nodes.data_ptr<int>()), off.data_ptr<int>(), T,
+        feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), B, acc.data_ptr<double>());
+  } else {
+    hipLaunchKernelGGL((forest_kernel<false>), dim3(row_blocks, chunks),
+        dim3(BLOCK), 0, stream,
+        codes.data_ptr<short>(), nums.data_ptr<float>(), medians.data_ptr<float>(),
+        reinterpret_cast<const int4*>(nodes.data_ptr<int>()), off.data_ptr<int>(), T,
+        feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), B, acc.data_ptr<double>());
+  }
+  HIP_CHECK(hipGetLastError());
+  return acc;
+}
+
 std::vector<torch::Tensor> drift_stats(
     torch::Tensor codes, torch::Tensor nums, torch::Tensor medians,
     torch::Tensor ref_sorted, torch::Tensor rs_off, torch::Tensor cat_off,
@@ -377,7 +486,7 @@ std::vector<torch::Tensor> drift_stats(
   int m_pow2 = 1;
   while (m_pow2 < B) m_pow2 <<= 1;
   m_pow2 = std::max(m_pow2, 2);
-  hipLaunchKernelGGL(ks_kernel, dim3(N_NUM), dim3(BLOCK),
+  hipLaunchKernelGGL((ks_kernel_t<256>), dim3(N_NUM), dim3(BLOCK),
       (size_t)m_pow2 * sizeof(float), stream,
       nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, B, m_pow2,
       /*ref_lds=*/0, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
@@ -538,7 +647,7 @@ torch::Tensor ks_stats(
   const int F = (int)nums.size(1);
   TORCH_CHECK(B <= MAX_DRIFT_ROWS, "K-S batch too large: ", B);
   TORCH_CHECK((int)rs_off.size(0) == F + 1, "rs_off size mismatch");
-  TORCH_CHECK(block == 256, "ks_kernel is __launch_bounds__(256)");
+  TORCH_CHECK(block == 256 || block == 512, "block must be 256 or 512");
   auto ks_d = torch::empty({F},
       torch::TensorOptions().dtype(torch::kFloat32).device(nums.device()));
   hipStream_t stream = c10::hip::getCurrentHIPStream();
@@ -553,10 +662,16 @@ torch::Tensor ks_stats(
     smem += (size_t)maxr * sizeof(float);
     TORCH_CHECK(smem <= KS_LDS_BYTES_MAX, "ref too large for LDS");
   }
-  hipLaunchKernelGGL(ks_kernel, dim3(F), dim3((int)block), smem, stream,
-      nums.data_ptr<float>(), medians.data_ptr<float>(), F, B, m_pow2,
-      (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
-      ks_d.data_ptr<float>());
+  if (block == 512)
+    hipLaunchKernelGGL((ks_kernel_t<512>), dim3(F), dim3(512), smem, stream,
+        nums.data_ptr<float>(), medians.data_ptr<float>(), F, B, m_pow2,
+        (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
+        ks_d.data_ptr<float>());
+  else
+    hipLaunchKernelGGL((ks_kernel_t<256>), dim3(F), dim3(256), smem, stream,
+        nums.data_ptr<float>(), medians.data_ptr<float>(), F, B, m_pow2,
+        (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
+        ks_d.data_ptr<float>());
   HIP_CHECK(hipGetLastError());
   return ks_d;
 }
@@ -628,7 +743,7 @@ struct ScoreSession {
       max_ref_len = std::max(max_ref_len, (int64_t)(rs_acc[j + 1] - rs_acc[j]));
 
     // opt in to >64 KiB dynamic LDS for the LDS-staged K-S path
-    (void)hipFuncSetAttribute(reinterpret_cast<const void*>(ks_kernel),
+    (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&ks_kernel_t<256>),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)KS_LDS_BYTES);
 
     HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
@@ -708,7 +823,7 @@ struct ScoreSession {
       // ks_kernel is __launch_bounds__(256), so block stays 256.
       (void)ref_bytes;
       const bool ref_lds = false;
-      hipLaunchKernelGGL(ks_kernel, dim3(N_NUM), dim3(BLOCK),
+      hipLaunchKernelGGL((ks_kernel_t<256>), dim3(N_NUM), dim3(BLOCK),
           batch_bytes + (ref_lds ? ref_bytes : 0), stream2,
           d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b, m_pow2,
           (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
@@ -1387,6 +1502,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Fused impute + logistic linear score + robust-z outlier (gfx950)");
   m.def("ks_stats_sorted", &ks_stats_sorted,
         "Exact K-S D per column for pre-sorted batch columns (gfx950)");
+  m.def("forest_ilp_bench", &forest_ilp_bench,
+        "A/B: forest traversal, 1 vs 2 trees per thread");
   m.def("ks_stats", &ks_stats,
         "Exact two-sample K-S D per column, any column count (gfx950)",
         py::arg("nums"), py::arg("medians"), py::arg("ref_sorted"),
