@@ -161,13 +161,17 @@ def train_dense(
     for f0 in range(0, n_feats, chunk):
         f1 = min(f0 + chunk, n_feats)
         r = torch.randn(ref_rows, f1 - f0, generator=rgen, device=dev)
-        s = r.sort(dim=0).values  # [n_ref, chunk]
-        ref_sorted[f0:f1] = s.t()
-        med[f0:f1] = s[ref_rows // 2]
-        q1 = s[int(0.25 * (ref_rows - 1))]
-        q3 = s[int(0.75 * (ref_rows - 1))]
-        iqr[f0:f1] = (q3 - q1).clamp_min(1e-6)
-        del r, s
+        for k in range(f1 - f0):
+            # per-column 1D radix sort (the segmented dim-0 sort faults on
+            # 10M-row segments in rocPRIM as shipped here)
+            s = r[:, k].contiguous().sort().values
+            ref_sorted[f0 + k] = s
+            med[f0 + k] = s[ref_rows // 2]
+            iqr[f0 + k] = (
+                s[int(0.75 * (ref_rows - 1))] - s[int(0.25 * (ref_rows - 1))]
+            ).clamp_min(1e-6)
+            del s
+        del r
 
     model = DenseModel(
         weight=w.float(),

@@ -782,16 +782,19 @@ struct ScoreSession {
     HIP_CHECK(hipMemsetAsync(acc_cls, 0, (size_t)(2 * b) * sizeof(double), stream));
 
     const int row_blocks = ceil_div(b, BLOCK);
+    // 2-tree-ILP traversal measured faster at every batch size
+    // (bench/kernel_micro.py: -12% @1k rows, -42% @16k); grid-y covers
+    // ceil(T/2) chunks, each thread walking trees t and t+gridDim.y.
     auto chunks = [&](int64_t t) {
-      return std::max(1, std::min(ceil_div(2048, row_blocks), (int)t));
+      return std::max(1, std::min(ceil_div(2048, row_blocks), (int)((t + 1) / 2)));
     };
-    hipLaunchKernelGGL((forest_kernel<false>), dim3(row_blocks, chunks(t_cls)),
+    hipLaunchKernelGGL((forest_kernel_ilp<false>), dim3(row_blocks, chunks(t_cls)),
         dim3(BLOCK), 0, stream,
         d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
         reinterpret_cast<const int4*>(cls_nodes.data_ptr<int>()),
         cls_off.data_ptr<int>(), (int)t_cls,
         feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), b, acc_cls);
-    hipLaunchKernelGGL((forest_kernel<true>), dim3(row_blocks, chunks(t_if)),
+    hipLaunchKernelGGL((forest_kernel_ilp<true>), dim3(row_blocks, chunks(t_if)),
         dim3(BLOCK), 0, stream,
         d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
         reinterpret_cast<const int4*>(if_nodes.data_ptr<int>()),
@@ -823,7 +826,9 @@ struct ScoreSession {
       // ks_kernel is __launch_bounds__(256), so block stays 256.
       (void)ref_bytes;
       const bool ref_lds = false;
-      hipLaunchKernelGGL((ks_kernel_t<256>), dim3(N_NUM), dim3(BLOCK),
+      // 512 threads: halves the bitonic's serial depth per thread
+      // (kernel_micro: 48.6->31.2 us @1k, 963->509 @16k)
+      hipLaunchKernelGGL((ks_kernel_t<512>), dim3(N_NUM), dim3(512),
           batch_bytes + (ref_lds ? ref_bytes : 0), stream2,
           d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b, m_pow2,
           (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
