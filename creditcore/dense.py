@@ -224,7 +224,8 @@ class DenseEngine:
             return torch.from_numpy(np.ascontiguousarray(a)).to(dev, dtype)
 
         F = m.n_features
-        rs_off = np.arange(F + 1, dtype=np.int32) * m.n_ref
+        # int64: offsets exceed 2^31 at HBM-scale references (10M x 1k)
+        rs_off = np.arange(F + 1, dtype=np.int64) * m.n_ref
         self._gpu = {
             "torch": torch,
             "ext": ext,

@@ -259,7 +259,8 @@ __global__ __launch_bounds__(BS) void ks_kernel_t(
     int m_pow2,                           // next pow2 >= n_rows
     int ref_lds,                          // 1 => stage ref column in LDS
     const float* __restrict__ ref_sorted, // concatenated per-feature refs
-    const int* __restrict__ rs_off,       // [n_cols+1]
+    const long long* __restrict__ rs_off, // [n_cols+1] (i64: offsets exceed
+                                          // 2^31 for HBM-scale references)
     float* __restrict__ ks_d)             // [n_cols]
 {
   extern __shared__ float s_vals[];  // [m_pow2] batch | [n_ref] staged ref
@@ -274,8 +275,8 @@ __global__ __launch_bounds__(BS) void ks_kernel_t(
     }
     s_vals[i] = v;
   }
-  const int ref_lo = rs_off[j];
-  const int n_ref_j = rs_off[j + 1] - ref_lo;
+  const long long ref_lo = rs_off[j];
+  const int n_ref_j = (int)(rs_off[j + 1] - ref_lo);
   if (ref_lds) {
     float* s_ref = s_vals + m_pow2;
     for (int i = threadIdx.x; i < n_ref_j; i += blockDim.x)
@@ -474,6 +475,7 @@ std::vector<torch::Tensor> drift_stats(
       torch::TensorOptions().dtype(torch::kInt32).device(codes.device()));
   auto ks_d = torch::empty({N_NUM},
       torch::TensorOptions().dtype(torch::kFloat32).device(codes.device()));
+  auto rs_off64 = rs_off.to(torch::kInt64);
 
   hipStream_t stream = c10::hip::getCurrentHIPStream();
 
@@ -489,8 +491,8 @@ std::vector<torch::Tensor> drift_stats(
   hipLaunchKernelGGL((ks_kernel_t<256>), dim3(N_NUM), dim3(BLOCK),
       (size_t)m_pow2 * sizeof(float), stream,
       nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, B, m_pow2,
-      /*ref_lds=*/0, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
-      ks_d.data_ptr<float>());
+      /*ref_lds=*/0, ref_sorted.data_ptr<float>(),
+      rs_off64.data_ptr<int64_t>(), ks_d.data_ptr<float>());
   HIP_CHECK(hipGetLastError());
 
   return {hist, ks_d};
@@ -579,14 +581,14 @@ __global__ __launch_bounds__(BLOCK) void ks_scan_kernel(
     int n_cols,
     int n_rows,
     const float* __restrict__ ref_sorted,
-    const int* __restrict__ rs_off,
+    const long long* __restrict__ rs_off,
     unsigned int* __restrict__ ks_bits)   // [F] f32 bits, pre-zeroed
 {
   const int j = blockIdx.x;
   const int i = blockIdx.y * blockDim.x + threadIdx.x;
   const int m = n_rows;
-  const int lo = rs_off[j];
-  const int n = rs_off[j + 1] - lo;
+  const long long lo = rs_off[j];
+  const int n = (int)(rs_off[j + 1] - lo);
   const float* __restrict__ ref = ref_sorted + lo;
   const float* __restrict__ col = xs + (size_t)j * n_rows;  // contiguous
 
@@ -627,9 +629,10 @@ torch::Tensor ks_stats_sorted(
   auto bits = torch::zeros({F},
       torch::TensorOptions().dtype(torch::kInt32).device(xs_sorted.device()));
   hipStream_t stream = c10::hip::getCurrentHIPStream();
+  auto rs_off64 = rs_off.to(torch::kInt64);
   hipLaunchKernelGGL(ks_scan_kernel, dim3(F, ceil_div(B, BLOCK)), dim3(BLOCK), 0, stream,
       xs_sorted.data_ptr<float>(), F, B,
-      ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
+      ref_sorted.data_ptr<float>(), rs_off64.data_ptr<int64_t>(),
       reinterpret_cast<unsigned int*>(bits.data_ptr<int>()));
   HIP_CHECK(hipGetLastError());
   return bits.view(torch::kFloat32);
@@ -654,10 +657,11 @@ torch::Tensor ks_stats(
   int m_pow2 = 2;
   while (m_pow2 < B) m_pow2 <<= 1;
   size_t smem = (size_t)m_pow2 * sizeof(float);
+  auto rs_off64 = rs_off.to(torch::kInt64);
   if (ref_lds) {
     int64_t maxr = 0;
-    auto ro = rs_off.cpu();
-    auto* rp = ro.data_ptr<int>();
+    auto ro = rs_off64.cpu();
+    auto* rp = ro.data_ptr<int64_t>();
     for (int j = 0; j < F; ++j) maxr = std::max<int64_t>(maxr, rp[j + 1] - rp[j]);
     smem += (size_t)maxr * sizeof(float);
     TORCH_CHECK(smem <= KS_LDS_BYTES_MAX, "ref too large for LDS");
@@ -665,13 +669,13 @@ torch::Tensor ks_stats(
   if (block == 512)
     hipLaunchKernelGGL((ks_kernel_t<512>), dim3(F), dim3(512), smem, stream,
         nums.data_ptr<float>(), medians.data_ptr<float>(), F, B, m_pow2,
-        (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
-        ks_d.data_ptr<float>());
+        (int)ref_lds, ref_sorted.data_ptr<float>(),
+        rs_off64.data_ptr<int64_t>(), ks_d.data_ptr<float>());
   else
     hipLaunchKernelGGL((ks_kernel_t<256>), dim3(F), dim3(256), smem, stream,
         nums.data_ptr<float>(), medians.data_ptr<float>(), F, B, m_pow2,
-        (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
-        ks_d.data_ptr<float>());
+        (int)ref_lds, ref_sorted.data_ptr<float>(),
+        rs_off64.data_ptr<int64_t>(), ks_d.data_ptr<float>());
   HIP_CHECK(hipGetLastError());
   return ks_d;
 }
@@ -714,7 +718,8 @@ struct ScoreSession {
     if_nodes = up_i32("if_nodes");
     if_off = up_i32("if_tree_offsets");
     ref_sorted = up_f32("ref_sorted");
-    rs_off = up_i32("ref_sorted_offsets");
+    rs_off = py::cast<torch::Tensor>(model["ref_sorted_offsets"])
+                 .to(devopt.dtype(torch::kInt64)).contiguous();
     cat_off = up_i32("ref_cat_offsets");
     if_denom = py::cast<double>(model["if_denom"]);
     if_offset = py::cast<double>(model["if_offset"]);
@@ -831,8 +836,8 @@ struct ScoreSession {
       hipLaunchKernelGGL((ks_kernel_t<512>), dim3(N_NUM), dim3(512),
           batch_bytes + (ref_lds ? ref_bytes : 0), stream2,
           d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b, m_pow2,
-          (int)ref_lds, ref_sorted.data_ptr<float>(), rs_off.data_ptr<int>(),
-          ksd.data_ptr<float>());
+          (int)ref_lds, ref_sorted.data_ptr<float>(),
+          rs_off.data_ptr<int64_t>(), ksd.data_ptr<float>());
       HIP_CHECK(hipMemcpyAsync(pin_hist.data_ptr(), hist.data_ptr(),
           (size_t)total_bins * sizeof(int), hipMemcpyDeviceToHost, stream2));
       HIP_CHECK(hipMemcpyAsync(pin_ksd.data_ptr(), ksd.data_ptr(),
