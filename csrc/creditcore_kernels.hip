@@ -259,7 +259,7 @@ __global__ __launch_bounds__(BS) void ks_kernel_t(
     int m_pow2,                           // next pow2 >= n_rows
     int ref_lds,                          // 1 => stage ref column in LDS
     const float* __restrict__ ref_sorted, // concatenated per-feature refs
-    const long long* __restrict__ rs_off, // [n_cols+1] (i64: offsets exceed
+    const int64_t* __restrict__ rs_off,   // [n_cols+1] (i64: offsets exceed
                                           // 2^31 for HBM-scale references)
     float* __restrict__ ks_d)             // [n_cols]
 {
@@ -275,7 +275,7 @@ __global__ __launch_bounds__(BS) void ks_kernel_t(
     }
     s_vals[i] = v;
   }
-  const long long ref_lo = rs_off[j];
+  const int64_t ref_lo = rs_off[j];
   const int n_ref_j = (int)(rs_off[j + 1] - ref_lo);
   if (ref_lds) {
     float* s_ref = s_vals + m_pow2;
@@ -581,13 +581,13 @@ __global__ __launch_bounds__(BLOCK) void ks_scan_kernel(
     int n_cols,
     int n_rows,
     const float* __restrict__ ref_sorted,
-    const long long* __restrict__ rs_off,
+    const int64_t* __restrict__ rs_off,
     unsigned int* __restrict__ ks_bits)   // [F] f32 bits, pre-zeroed
 {
   const int j = blockIdx.x;
   const int i = blockIdx.y * blockDim.x + threadIdx.x;
   const int m = n_rows;
-  const long long lo = rs_off[j];
+  const int64_t lo = rs_off[j];
   const int n = (int)(rs_off[j + 1] - lo);
   const float* __restrict__ ref = ref_sorted + lo;
   const float* __restrict__ col = xs + (size_t)j * n_rows;  // contiguous
