@@ -53,6 +53,8 @@ def main():
         keep_on_device=(device == "cuda"),
         log=lambda *x: print(*x, file=sys.stderr),
     )
+    if device == "cuda":
+        torch.cuda.synchronize()  # train queues async work (ref-build sorts)
     train_s = time.time() - t0
     eng = DenseEngine(model, device=device)
 
