@@ -37,8 +37,10 @@ class MicroBatcher:
         score_arrays,  # callable(codes, nums) -> dict (engine.score_arrays)
         max_rows: int = 8192,
         max_wait_us: int = 300,
+        score_single=None,  # optional fast path for single-request flushes
     ):
         self.score_arrays = score_arrays
+        self.score_single = score_single
         self.max_rows = int(max_rows)
         self.max_wait = max(0.0, max_wait_us * 1e-6)
         self._pending: list[_Pending] = []
@@ -100,6 +102,22 @@ class MicroBatcher:
             await self._flush_one(batch)
 
     async def _flush_one(self, batch):
+        if len(batch) == 1 and self.score_single is not None:
+            # no merge happened: the whole request can take the engine's
+            # single-call wire-out fast path
+            p = batch[0]
+            loop = asyncio.get_running_loop()
+            try:
+                out = await loop.run_in_executor(
+                    None, self.score_single, p.codes, p.nums
+                )
+            except Exception as e:
+                if not p.future.done():
+                    p.future.set_exception(e)
+                return
+            if not p.future.done():
+                p.future.set_result(out)
+            return
         codes = np.concatenate([p.codes for p in batch], axis=0)
         nums = np.concatenate([p.nums for p in batch], axis=0)
         loop = asyncio.get_running_loop()

@@ -215,7 +215,12 @@ class ScoringEngine:
         resp = g["ext"].build_response_json(
             g["sess"].pin_outs, b, np.ascontiguousarray(pvals), FEATURES
         )
-        return {"response_bytes": resp, "rows": b}
+        return {
+            "response_bytes": resp,
+            "rows": b,
+            # for node-global drift aggregation by the serving layer
+            "cat_hist": g["np_hist"].copy(),
+        }
 
     def score_records(self, records) -> dict:
         """Score a request body (list of dicts / DataFrame); returns the

@@ -96,21 +96,39 @@ async def lifespan(app: FastAPI):
     # the per-request drift stats feeding it come from the HIP kernels).
     drift_sync = DriftSync(engines[0].packed, device="cpu")
 
+    def _fold_drift(out, nums):
+        if "cat_hist" in out:
+            drift_sync.accumulate(out["cat_hist"], nums)
+            if drift_sync.batches % max(cfg.drift_sync_period, 1) == 0:
+                drift_sync.allreduce()
+                state["metrics"].observe_drift_sync()
+
     def scorer(e: ScoringEngine):
         def run(codes, nums):
             out = e.score_arrays(codes, nums)
-            if "cat_hist" in out:
-                drift_sync.accumulate(out["cat_hist"], nums)
-                if drift_sync.batches % max(cfg.drift_sync_period, 1) == 0:
-                    drift_sync.allreduce()
-                    state["metrics"].observe_drift_sync()
+            _fold_drift(out, nums)
+            return out
+
+        return run
+
+    def scorer_single(e: ScoringEngine):
+        # GPU engines serialize the wire response in C for solo flushes
+        if e.device != "cuda":
+            return None
+
+        def run(codes, nums):
+            out = e.score_encoded_bytes(codes, nums)
+            _fold_drift(out, nums)
             return out
 
         return run
 
     batchers = [
         MicroBatcher(
-            scorer(e), max_rows=cfg.max_batch_rows, max_wait_us=cfg.batch_wait_us
+            scorer(e),
+            max_rows=cfg.max_batch_rows,
+            max_wait_us=cfg.batch_wait_us,
+            score_single=scorer_single(e),
         )
         for e in engines
     ]
@@ -190,6 +208,19 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
             raise HTTPException(status_code=500, detail=f"scoring failed: {e}")
         latency_ms = (time.perf_counter() - t0) * 1e3
 
+        if "response_bytes" in out:  # solo-flush wire-out fast path
+            rb = out["response_bytes"]
+            metrics.observe_request(out["rows"], latency_ms)
+            reqlog.log_model_output_raw(
+                cfg.service_name,
+                request_id,
+                rb.decode("utf-8", "replace"),
+                latency_ms=latency_ms,
+                rows=out["rows"],
+                device=f"{engines[idx].device}:{engines[idx].device_index}",
+            )
+            return rb
+
         one_minus = (
             np.float32(1.0) - np.asarray(out["p_vals"], dtype=np.float32)
         ).astype(np.float64)
@@ -225,8 +256,10 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         }
     }
 
-    def _json_response(payload: dict) -> Response:
+    def _json_response(payload) -> Response:
         # shape is correct by construction; skip response-model revalidation
+        if isinstance(payload, (bytes, str)):
+            return Response(payload, media_type="application/json")
         return Response(json.dumps(payload), media_type="application/json")
 
     @app.post("/predict", response_model=ModelOutput, openapi_extra=_openapi_extra)

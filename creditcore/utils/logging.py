@@ -30,6 +30,28 @@ def log_inference_data(service_name: str, request_id: str, data_json: str) -> No
     )
 
 
+def log_model_output_raw(
+    service_name: str,
+    request_id: str,
+    response_json: str,
+    latency_ms: float | None = None,
+    rows: int | None = None,
+    device: str | None = None,
+) -> None:
+    """Like log_model_output but embeds an already-serialized response JSON
+    verbatim (the serving wire-out fast path)."""
+    logger.info(
+        '{"service_name": %s, "type": "ModelOutput", "request_id": "%s", '
+        '"latency_ms": %.3f, "rows": %s, "device": "%s", "data": %s}',
+        json.dumps(service_name),
+        request_id,
+        latency_ms if latency_ms is not None else -1.0,
+        rows if rows is not None else "null",
+        device or "",
+        response_json,
+    )
+
+
 def log_model_output(
     service_name: str,
     request_id: str,

@@ -94,6 +94,34 @@ def test_engine_refuses_silent_fallback(packed, monkeypatch):
         ScoringEngine(packed, device="cuda", device_index=0)
 
 
+def test_serving_stack_on_gpu(model_dir):
+    """Whole serving stack on the GPU engine (TestClient): contract-valid
+    responses through both the solo-flush bytes path and the merged path."""
+    from fastapi.testclient import TestClient
+
+    from creditcore.config import ServeConfig
+    from creditcore.schema import SAMPLE_REQUEST, ModelOutput
+    from creditcore.serve import create_app
+    from creditcore.data import make_request_batch
+
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cuda"
+    app = create_app(cfg)
+    with TestClient(app) as client:
+        r = client.post("/predict", json=SAMPLE_REQUEST)
+        assert r.status_code == 200
+        ModelOutput.model_validate(r.json())
+        r = client.post("/score", json=make_request_batch(300, seed=2))
+        assert r.status_code == 200
+        body = r.json()
+        assert len(body["predictions"]) == 300
+        assert all(0.0 <= p <= 1.0 for p in body["predictions"])
+        assert client.get("/healthz").json()["status"] == "ok"
+        drift = client.get("/drift").json()
+        assert drift["rows"] >= 301
+
+
 def test_native_code_is_loaded(gpu_engine):
     """The loaded extension must be the in-tree .so (native-code check)."""
     import creditcore._ccore as ccore
