@@ -82,8 +82,9 @@ def main():
             )
         )
         results[f"forest_pipeline b={b}"] = round(us, 2)
-        # A/B: 1 vs 2 trees per thread (classifier forest alone)
-        for ilp in (0, 1):
+        # A/B: ILP depth x grid layout (classifier forest alone)
+        # 0=1-tree, 1=ilp2, 2=ilp4, 3=ilp2+XCD-transposed grid, 4=ilp4+transposed
+        for ilp in (0, 1, 2, 3, 4):
             us = timed(
                 lambda: ext.forest_ilp_bench(
                     d_codes, d_nums, cls_nodes, cls_off, fc, fk, medians, ilp
@@ -91,8 +92,9 @@ def main():
             )
             results[f"forest_cls b={b} ilp={ilp}"] = round(us, 2)
         a = ext.forest_ilp_bench(d_codes, d_nums, cls_nodes, cls_off, fc, fk, medians, 0)
-        c = ext.forest_ilp_bench(d_codes, d_nums, cls_nodes, cls_off, fc, fk, medians, 1)
-        results[f"forest_ilp_parity b={b}"] = float((a - c).abs().max().item())
+        for v in (1, 2, 3, 4):
+            c = ext.forest_ilp_bench(d_codes, d_nums, cls_nodes, cls_off, fc, fk, medians, v)
+            results[f"forest_ilp_parity b={b} v={v}"] = float((a - c).abs().max().item())
 
     print(json.dumps(results, indent=2))
 

@@ -184,6 +184,84 @@ __global__ __launch_bounds__(BLOCK) void forest_kernel_ilp(
   atomicAdd(&acc[row], local);
 }
 
+// 4-tree ILP + optional transposed grid (blockIdx.x = tree chunk so the
+// dispatcher's id%8 XCD placement gives adjacent chunks to different XCDs,
+// keeping each XCD's L2 on a tree subset). Experimental A/B variants.
+template <bool DIRECT, int ILP, bool SWAP_GRID>
+__global__ __launch_bounds__(BLOCK) void forest_kernel_ilpN(
+    const short* __restrict__ codes,
+    const float* __restrict__ nums,
+    const float* __restrict__ medians,
+    const int4* __restrict__ nodes,
+    const int* __restrict__ tree_off,
+    int n_trees,
+    const int* __restrict__ feat_col,
+    const int* __restrict__ feat_code,
+    int n_rows,
+    double* __restrict__ acc)
+{
+  __shared__ short s_codes[N_CAT * BLOCK];
+  __shared__ float s_nums[N_NUM * BLOCK];
+  const int tid = threadIdx.x;
+  const int row_blk = SWAP_GRID ? blockIdx.y : blockIdx.x;
+  const int chunk = SWAP_GRID ? blockIdx.x : blockIdx.y;
+  const int n_chunks = SWAP_GRID ? gridDim.x : gridDim.y;
+  const int row = row_blk * BLOCK + tid;
+  if (row < n_rows) {
+    if (!DIRECT) {
+#pragma unroll
+      for (int c = 0; c < N_CAT; ++c) s_codes[c * BLOCK + tid] = codes[row * N_CAT + c];
+    }
+#pragma unroll
+    for (int c = 0; c < N_NUM; ++c) {
+      const float v = nums[row * N_NUM + c];
+      s_nums[c * BLOCK + tid] = isnan(v) ? medians[c] : v;
+    }
+  }
+  if (row >= n_rows) return;
+
+  auto value_of = [&](int f) -> float {
+    if (DIRECT) return s_nums[f * BLOCK + tid];
+    const int col = feat_col[f];
+    const int code = feat_code[f];
+    return (code >= 0) ? ((s_codes[col * BLOCK + tid] == (short)code) ? 1.0f : 0.0f)
+                       : s_nums[col * BLOCK + tid];
+  };
+
+  double local = 0.0;
+  for (int t = chunk; t < n_trees; t += ILP * n_chunks) {
+    int4 nd[ILP];
+    int base[ILP];
+    bool live[ILP];
+#pragma unroll
+    for (int k = 0; k < ILP; ++k) {
+      const int tk = t + k * n_chunks;
+      live[k] = tk < n_trees;
+      if (live[k]) {
+        base[k] = tree_off[tk];
+        nd[k] = nodes[base[k]];
+      }
+    }
+    bool any = true;
+    while (any) {
+      any = false;
+#pragma unroll
+      for (int k = 0; k < ILP; ++k) {
+        if (!live[k]) continue;
+        if (nd[k].x >= 0) {
+          const float v = value_of(nd[k].x);
+          nd[k] = nodes[base[k] + ((v <= __int_as_float(nd[k].y)) ? nd[k].z : nd[k].w)];
+          any = true;
+        } else {
+          local += (double)__int_as_float(nd[k].y);
+          live[k] = false;
+        }
+      }
+    }
+  }
+  atomicAdd(&acc[row], local);
+}
+
 __global__ __launch_bounds__(BLOCK) void finalize_kernel(
     const double* __restrict__ cls_acc,
     const double* __restrict__ if_acc,
@@ -444,12 +522,23 @@ torch::Tensor forest_ilp_bench(
   const int row_blocks = ceil_div(B, BLOCK);
   int chunks = std::max(1, std::min(ceil_div(2048, row_blocks), T));
   if (use_ilp) {
-    chunks = std::max(1, std::min(ceil_div(2048, row_blocks), (T + 1) / 2));
-    hipLaunchKernelGGL((forest_kernel_ilp<false>), dim3(row_blocks, chunks),
-        dim3(BLOCK), 0, stream,
+    const int ilp = (use_ilp == 2 || use_ilp == 4) ? 4 : 2;
+    const bool swap = use_ilp >= 3;
+    chunks = std::max(1, std::min(ceil_div(2048, row_blocks), (T + ilp - 1) / ilp));
+    dim3 grid = swap ? dim3(chunks, row_blocks) : dim3(row_blocks, chunks);
+    auto args = std::make_tuple(
         codes.data_ptr<short>(), nums.data_ptr<float>(), medians.data_ptr<float>(),
         reinterpret_cast<const int4*>(nodes.data_ptr<int>()), off.data_ptr<int>(), T,
         feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), B, acc.data_ptr<double>());
+    auto launch = [&](auto kernel) {
+      std::apply([&](auto... a) {
+        hipLaunchKernelGGL(kernel, grid, dim3(BLOCK), 0, stream, a...);
+      }, args);
+    };
+    if (use_ilp == 1) launch(forest_kernel_ilpN<false, 2, false>);
+    else if (use_ilp == 2) launch(forest_kernel_ilpN<false, 4, false>);
+    else if (use_ilp == 3) launch(forest_kernel_ilpN<false, 2, true>);
+    else launch(forest_kernel_ilpN<false, 4, true>);
   } else {
     hipLaunchKernelGGL((forest_kernel<false>), dim3(row_blocks, chunks),
         dim3(BLOCK), 0, stream,
