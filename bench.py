@@ -107,6 +107,15 @@ def main():
 
     engine = ScoringEngine(packed, device=device, device_index=local_rank)
 
+    # Node-global drift state (BASELINE config 4): per-replica histograms,
+    # RCCL all-reduce over xGMI every sync period — exercised inside the
+    # timed loop when running distributed.
+    from creditcore.parallel import DriftSync
+
+    drift_dev = f"cuda:{local_rank}" if (distributed and device == "cuda") else "cpu"
+    drift_sync = DriftSync(packed, device=drift_dev) if distributed else None
+    DRIFT_SYNC_PERIOD = 32
+
     # Pre-generate a pool of request bodies in the wire format (JSON bytes);
     # each timed step runs the full serving compute path: native JSON parse
     # + encode -> pinned H2D -> HIP kernels -> D2H -> drift p-values ->
@@ -124,7 +133,18 @@ def main():
             assert out["rows"] == args.rows
             return out
 
-        run_steps = lambda k: [one_step(i) for i in range(k)]  # noqa: E731
+        def run_steps(k: int, step_times=None):
+            outs = []
+            t_prev = time.perf_counter()
+            for i in range(k):
+                outs.append(one_step(i))
+                if step_times is not None:
+                    t_now = time.perf_counter()
+                    step_times.append(t_now - t_prev)
+                    t_prev = t_now
+            return outs
+
+        drift_sync = None
     else:
         # Steady-state serving pipeline: request i+1's JSON parse (GIL
         # released in the C parser) overlaps request i's GPU work. Every
@@ -136,12 +156,13 @@ def main():
         executor = ThreadPoolExecutor(max_workers=2)
         DEPTH = 3  # encode i+1..i+3 overlap scoring of i (C parser drops the GIL)
 
-        def run_steps(k: int):
+        def run_steps(k: int, step_times=None):
             outs = []
             q = deque(
                 executor.submit(engine.encode_json_body, pool[i % len(pool)])
                 for i in range(min(DEPTH, k))
             )
+            t_prev = time.perf_counter()
             for i in range(k):
                 codes, nums = q.popleft().result()
                 nxt = i + DEPTH
@@ -151,7 +172,21 @@ def main():
                     )
                 out = engine.score_encoded_bytes(codes, nums)
                 assert out["rows"] == args.rows
+                if (
+                    drift_sync is not None
+                    and "cat_hist" in out
+                    and (i + 1) % DRIFT_SYNC_PERIOD == 0
+                ):
+                    drift_sync.accumulate(
+                        torch.from_numpy(out["cat_hist"]).to(drift_dev),
+                        torch.from_numpy(nums).to(drift_dev),
+                    )
+                    drift_sync.allreduce()
                 outs.append(out)
+                if step_times is not None:
+                    t_now = time.perf_counter()
+                    step_times.append(t_now - t_prev)
+                    t_prev = t_now
             return outs
 
     run_steps(args.warmup)
@@ -167,8 +202,9 @@ def main():
             torch.cuda.synchronize()
 
     sync()
+    step_times: list = []
     t0 = time.perf_counter()
-    run_steps(args.steps)
+    run_steps(args.steps, step_times)
     if device == "cuda":
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
@@ -214,6 +250,11 @@ def main():
                         "rows_per_sec": round(requests_per_sec * args.rows, 1),
                         "with_drift": with_drift,
                         "train_rows": TRAIN_ROWS,
+                        "p50_ms": round(sorted(step_times)[len(step_times) // 2] * 1e3, 4),
+                        "p99_ms": round(
+                            sorted(step_times)[min(len(step_times) - 1, int(0.99 * len(step_times)))] * 1e3,
+                            4,
+                        ),
                         "parallelism": f"dp{n_gpus}",
                         "device": device,
                     },
