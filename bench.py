@@ -81,6 +81,9 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world_size > 1
+    if device == "cuda":
+        # tolerate more ranks than GPUs (e.g. RCCL smoke tests on one GPU)
+        local_rank = local_rank % max(torch.cuda.device_count(), 1)
     if distributed:
         import torch.distributed as dist
 
