@@ -21,8 +21,6 @@ import os
 import sys
 import time
 
-import numpy as np
-
 REQUEST_ROWS = 1024
 BENCH_MODEL = {"n_estimators": 500, "max_depth": 16, "criterion": "gini"}
 TRAIN_ROWS = 20_000

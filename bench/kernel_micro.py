@@ -9,7 +9,6 @@ session picks measured-best settings:
 from __future__ import annotations
 
 import json
-import time
 
 import numpy as np
 

@@ -11,7 +11,6 @@ explicitly via device="cpu").
 from __future__ import annotations
 
 import importlib
-import os
 
 _ext = None
 _err: Exception | None = None

@@ -35,7 +35,6 @@ extra "unseen" bin.
 
 from __future__ import annotations
 
-import os
 from collections import deque
 from dataclasses import dataclass, field
 

@@ -21,7 +21,6 @@ Stages:
 
 from __future__ import annotations
 
-import json
 import os
 import shutil
 import subprocess

@@ -30,13 +30,12 @@ from datetime import datetime, timezone
 
 import cloudpickle
 import joblib
-import numpy as np
 import pandas as pd
 import yaml
 
 from .models.drift import TabularDriftDetector
 from .models.iforest import IForestDetector
-from .schema import CATEGORICAL_FEATURES, FEATURES, NUMERIC_FEATURES
+from .schema import CATEGORICAL_FEATURES, NUMERIC_FEATURES
 
 PYFUNC_ARTIFACTS_SUBDIR = "artifacts"
 CLASSIFIER_PKL = "classifier/model/model.pkl"  # nesting per 02-register cell-9

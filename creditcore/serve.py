@@ -30,14 +30,7 @@ from .config import ServeConfig
 from .batching import MicroBatcher
 from .engine import ScoringEngine, load_engine
 from .pack import encode_batch
-from .schema import (
-    CATEGORICAL_FEATURES,
-    FEATURES,
-    MISSING_CATEGORY,
-    NUMERIC_FEATURES,
-    LoanApplicant,
-    ModelOutput,
-)
+from .schema import FEATURES, LoanApplicant, ModelOutput
 from .utils import logging as reqlog
 from .utils.metrics import Metrics
 
