@@ -25,6 +25,9 @@ def _cmd_train(argv):
     p.add_argument("--n-rows", type=int, default=20_000)
     p.add_argument("--seed", type=int, default=2024)
     p.add_argument("--no-register", action="store_true")
+    p.add_argument("--algorithm", default="rf", choices=["rf", "gbt"],
+                   help="rf = the reference's RandomForest; gbt = "
+                        "gradient-boosted trees (same HIP traversal kernel)")
     p.add_argument("--data", default=None,
                    help="CSV with the UCI schema (the reference's curated "
                         "table, 00-create-external-table.ipynb); synthetic "
@@ -46,6 +49,7 @@ def _cmd_train(argv):
         seed=a.seed,
         register=not a.no_register,
         df=df,
+        algorithm=a.algorithm,
     )
     print(uri)
 

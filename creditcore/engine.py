@@ -64,6 +64,8 @@ class ScoringEngine:
             "ref_sorted": torch.from_numpy(p.ref_sorted),
             "ref_sorted_offsets": torch.from_numpy(p.ref_sorted_offsets),
             "ref_cat_offsets": torch.from_numpy(p.ref_cat_offsets),
+            "cls_kind": int(getattr(p, "cls_kind", 0)),
+            "cls_bias": float(getattr(p, "cls_bias", 0.0)),
             "if_denom": float(p.if_denom),
             "if_offset": float(p.if_offset),
             "if_threshold": float(p.if_threshold),

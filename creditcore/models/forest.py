@@ -25,12 +25,27 @@ from sklearn.preprocessing import OneHotEncoder
 from ..schema import CATEGORICAL_FEATURES, MISSING_CATEGORY, NUMERIC_FEATURES
 
 
-def make_classifier_pipeline(params: dict) -> Pipeline:
+def make_classifier_pipeline(params: dict, algorithm: str = "rf") -> Pipeline:
     """Build the classifier pipeline (reference 01-train cell-6).
 
-    ``params`` are RandomForestClassifier kwargs
-    (n_estimators / max_depth / criterion / random_state ...).
+    ``params`` are estimator kwargs (n_estimators / max_depth / random_state
+    ...). ``algorithm``: "rf" (the reference's RandomForest) or "gbt"
+    (GradientBoostingClassifier — the north star's "gradient-boosted-tree
+    traversal" family; same packed node-SoA format, scored by the same HIP
+    kernel with a sigmoid(sum + prior) finalize).
     """
+    if algorithm == "gbt":
+        from sklearn.ensemble import GradientBoostingClassifier
+
+        params = dict(params)
+        params.pop("criterion", None)  # rf-only knob from the search space
+        params.setdefault("n_estimators", 200)
+        estimator = GradientBoostingClassifier(**params)
+    elif algorithm == "rf":
+        estimator = RandomForestClassifier(**params, n_jobs=-1)
+    else:
+        raise ValueError(f"unknown algorithm: {algorithm}")
+
     categorical_transformer = Pipeline(
         steps=[
             (
@@ -50,6 +65,6 @@ def make_classifier_pipeline(params: dict) -> Pipeline:
     return Pipeline(
         [
             ("preprocessor", preprocessor),
-            ("classifier", RandomForestClassifier(**params, n_jobs=-1)),
+            ("classifier", estimator),
         ]
     )

@@ -73,6 +73,8 @@ def score_forest_cpu(
         return v
 
     s = _traverse_forest(packed.cls_nodes, packed.cls_tree_offsets, value_of, len(codes))
+    if getattr(packed, "cls_kind", 0) == 1:  # gradient-boosted: logit sum
+        return 1.0 / (1.0 + np.exp(-(s + packed.cls_bias)))
     return (s / packed.cls_n_trees).astype(np.float64)
 
 

@@ -147,6 +147,11 @@ class PackedModel:
     ref_cat_counts: np.ndarray  # i32[sum (vocab+1)]
     ref_cat_offsets: np.ndarray  # i32[N_CAT+1]
 
+    # classifier finalize: 0 = random forest (proba = leaf-fraction mean);
+    # 1 = gradient-boosted (proba = sigmoid(sum of lr-scaled leaves + prior))
+    cls_kind: int = 0
+    cls_bias: float = 0.0
+
     # optional linear scorer
     lin_weight: np.ndarray | None = None  # f32[F_total]
     lin_bias: float = 0.0
@@ -178,6 +183,8 @@ class PackedModel:
             n_onehot=self.n_onehot,
             cls_nodes=self.cls_nodes,
             cls_tree_offsets=self.cls_tree_offsets,
+            cls_kind=self.cls_kind,
+            cls_bias=self.cls_bias,
             if_nodes=self.if_nodes,
             if_tree_offsets=self.if_tree_offsets,
             if_denom=self.if_denom,
@@ -206,6 +213,8 @@ class PackedModel:
             n_onehot=int(z["n_onehot"]),
             cls_nodes=z["cls_nodes"],
             cls_tree_offsets=z["cls_tree_offsets"],
+            cls_kind=int(z["cls_kind"]) if "cls_kind" in z else 0,
+            cls_bias=float(z["cls_bias"]) if "cls_bias" in z else 0.0,
             if_nodes=z["if_nodes"],
             if_tree_offsets=z["if_tree_offsets"],
             if_denom=float(z["if_denom"]),
@@ -224,9 +233,12 @@ class PackedModel:
 
 def pack_classifier_pipeline(pipeline) -> dict:
     """Extract vocabularies, medians and the BFS node-SoA forest from the
-    sklearn pipeline built by make_classifier_pipeline."""
+    sklearn pipeline built by make_classifier_pipeline. Supports the
+    reference's RandomForestClassifier (leaf-fraction mean) and
+    GradientBoostingClassifier (lr-scaled regression leaves summed into a
+    logit on top of the prior)."""
     pre = pipeline.named_steps["preprocessor"]
-    rf = pipeline.named_steps["classifier"]
+    clf = pipeline.named_steps["classifier"]
 
     cat_pipe = pre.named_transformers_["categorical"]
     num_pipe = pre.named_transformers_["numeric"]
@@ -246,14 +258,34 @@ def pack_classifier_pipeline(pipeline) -> dict:
         feat_code.append(-1)
 
     trees = []
-    for est in rf.estimators_:
-        t = est.tree_
-        value = np.asarray(t.value, dtype=np.float64)  # (n_nodes, 1, 2)
-        sums = value[:, 0, :].sum(axis=1)
-        sums[sums == 0] = 1.0
-        leaf_p1 = value[:, 0, 1] / sums  # fraction of class 1 (normalised
-        # either way: sklearn >=1.4 already stores fractions)
-        trees.append(_pack_sklearn_tree(t, leaf_p1))
+    cls_kind, cls_bias = 0, 0.0
+    if hasattr(clf, "loss_") or type(clf).__name__ == "GradientBoostingClassifier":
+        cls_kind = 1
+        lr = float(clf.learning_rate)
+        # binary log-loss prior (raw score of the init estimator)
+        import sklearn.dummy
+
+        init = clf.init_
+        if isinstance(init, str) and init == "zero":
+            cls_bias = 0.0
+        elif isinstance(init, sklearn.dummy.DummyClassifier):
+            p1 = float(np.clip(init.class_prior_[1], 1e-12, 1 - 1e-12))
+            cls_bias = float(np.log(p1 / (1.0 - p1)))
+        else:  # pragma: no cover - custom init estimators
+            raise ValueError("unsupported GBT init estimator")
+        for est in clf.estimators_[:, 0]:
+            t = est.tree_
+            leaf_raw = np.asarray(t.value, dtype=np.float64)[:, 0, 0] * lr
+            trees.append(_pack_sklearn_tree(t, leaf_raw))
+    else:
+        for est in clf.estimators_:
+            t = est.tree_
+            value = np.asarray(t.value, dtype=np.float64)  # (n_nodes, 1, 2)
+            sums = value[:, 0, :].sum(axis=1)
+            sums[sums == 0] = 1.0
+            leaf_p1 = value[:, 0, 1] / sums  # fraction of class 1 (normalised
+            # either way: sklearn >=1.4 already stores fractions)
+            trees.append(_pack_sklearn_tree(t, leaf_p1))
     nodes, offsets = _concat_trees(trees)
 
     return {
@@ -264,6 +296,8 @@ def pack_classifier_pipeline(pipeline) -> dict:
         "n_onehot": n_onehot,
         "cls_nodes": nodes,
         "cls_tree_offsets": offsets,
+        "cls_kind": cls_kind,
+        "cls_bias": cls_bias,
     }
 
 

@@ -36,6 +36,8 @@ from .schema import FEATURES
 
 _META_FIELDS = [
     "n_onehot",
+    "cls_kind",
+    "cls_bias",
     "if_denom",
     "if_offset",
     "if_threshold",

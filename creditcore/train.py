@@ -56,7 +56,9 @@ class EvalRun:
     pipeline: object = field(repr=False, default=None)
 
 
-def _evaluate(params: dict, df: pd.DataFrame, seed: int = 2024) -> EvalRun:
+def _evaluate(
+    params: dict, df: pd.DataFrame, seed: int = 2024, algorithm: str = "rf"
+) -> EvalRun:
     """One hyperparameter evaluation (reference cell-7, minus the dead
     double-fits)."""
     df_train, df_test = train_test_split(
@@ -65,7 +67,7 @@ def _evaluate(params: dict, df: pd.DataFrame, seed: int = 2024) -> EvalRun:
     x_train, y_train = df_train[FEATURES], df_train[TARGET]
     x_test, y_test = df_test[FEATURES], df_test[TARGET]
 
-    estimator = make_classifier_pipeline({**params, "random_state": seed})
+    estimator = make_classifier_pipeline({**params, "random_state": seed}, algorithm)
     estimator.fit(x_train, y_train.values.ravel())
     y_pred = estimator.predict(x_test)
 
@@ -101,6 +103,7 @@ def train_model(
     seed: int = 2024,
     runs_dir: str | None = None,
     n_rows: int = 20_000,
+    algorithm: str = "rf",
 ) -> EvalRun:
     """Hyperparameter search; returns the best run by roc_auc
     (reference cell-8/10)."""
@@ -112,7 +115,7 @@ def train_model(
     for i in range(max_evals):
         incumbent = best.params if (best is not None and i >= n_startup) else None
         params = _sample_params(rng, incumbent)
-        run = _evaluate(params, df, seed=seed)
+        run = _evaluate(params, df, seed=seed, algorithm=algorithm)
         if runs_dir:
             rd = os.path.join(runs_dir, run.run_id)
             os.makedirs(rd, exist_ok=True)
@@ -155,13 +158,14 @@ def train_and_register(
     seed: int = 2024,
     df: pd.DataFrame | None = None,
     register: bool = True,
+    algorithm: str = "rf",
 ) -> str:
     """The full train -> package -> register job (the reference's 2-task
     Databricks DAG, train_register_model.yml:10-39). Returns the model URI
     (or the model dir when register=False)."""
     if df is None:
         df = make_uci_shaped_frame(n_rows=n_rows, seed=seed)
-    best = train_model(df=df, max_evals=max_evals, seed=seed)
+    best = train_model(df=df, max_evals=max_evals, seed=seed, algorithm=algorithm)
     drift, outlier = fit_detectors(df)
     registry.save_pyfunc_model(
         model_dir,

@@ -266,7 +266,9 @@ __global__ __launch_bounds__(BLOCK) void finalize_kernel(
     const double* __restrict__ cls_acc,
     const double* __restrict__ if_acc,
     int n_rows,
+    int cls_kind,        // 0 = RF leaf-fraction mean; 1 = GBT logit sum
     double inv_n_trees,
+    double cls_bias,
     double if_denom,
     double if_offset,
     double if_threshold,
@@ -276,7 +278,8 @@ __global__ __launch_bounds__(BLOCK) void finalize_kernel(
 {
   const int i = blockIdx.x * BLOCK + threadIdx.x;
   if (i >= n_rows) return;
-  proba[i] = cls_acc[i] * inv_n_trees;
+  proba[i] = cls_kind ? 1.0 / (1.0 + exp(-(cls_acc[i] + cls_bias)))
+                      : cls_acc[i] * inv_n_trees;
   // alibi instance_score = -(sklearn decision_function)
   //                      = 2^(-mean_depth_sum/denom) + offset_
   const double anomaly = exp2(-if_acc[i] / if_denom);
@@ -500,7 +503,7 @@ std::vector<torch::Tensor> score_forest_pipeline(
 
   hipLaunchKernelGGL(finalize_kernel, dim3(row_blocks), dim3(BLOCK), 0, stream,
       cls_acc.data_ptr<double>(), if_acc.data_ptr<double>(), B,
-      1.0 / (double)t_cls, if_denom, if_offset, if_threshold,
+      /*cls_kind=*/0, 1.0 / (double)t_cls, 0.0, if_denom, if_offset, if_threshold,
       proba.data_ptr<double>(), iscore.data_ptr<double>(), outlier.data_ptr<double>());
   HIP_CHECK(hipGetLastError());
 
@@ -784,6 +787,8 @@ struct ScoreSession {
   torch::Tensor d_codes, d_nums, acc, outs, hist, ksd;
   torch::Tensor pin_codes, pin_nums, pin_outs, pin_hist, pin_ksd;
   double if_denom{}, if_offset{}, if_threshold{};
+  int cls_kind{};
+  double cls_bias{};
   int64_t total_bins{}, t_cls{}, t_if{}, capacity{};
   int device_index{};
   hipStream_t stream{};
@@ -810,6 +815,8 @@ struct ScoreSession {
     rs_off = py::cast<torch::Tensor>(model["ref_sorted_offsets"])
                  .to(devopt.dtype(torch::kInt64)).contiguous();
     cat_off = up_i32("ref_cat_offsets");
+    if (model.contains("cls_kind")) cls_kind = py::cast<int>(model["cls_kind"]);
+    if (model.contains("cls_bias")) cls_bias = py::cast<double>(model["cls_bias"]);
     if_denom = py::cast<double>(model["if_denom"]);
     if_offset = py::cast<double>(model["if_offset"]);
     if_threshold = py::cast<double>(model["if_threshold"]);
@@ -896,7 +903,8 @@ struct ScoreSession {
 
     double* proba = outs.data_ptr<double>();
     hipLaunchKernelGGL(finalize_kernel, dim3(row_blocks), dim3(BLOCK), 0, stream,
-        acc_cls, acc_if, b, 1.0 / (double)t_cls, if_denom, if_offset, if_threshold,
+        acc_cls, acc_if, b, cls_kind, 1.0 / (double)t_cls, cls_bias,
+        if_denom, if_offset, if_threshold,
         proba, proba + b, proba + 2 * b);
 
     if (with_drift) {
