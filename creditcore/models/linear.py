@@ -1,14 +1,16 @@
 """Linear / logistic scorer — the "linear predict" model family.
 
 BASELINE.json's north star names "gradient-boosted-tree traversal / linear
-predict" as the tabular scoring hot path. The forest kernel covers tree
-traversal; this family covers dense linear scoring, whose GPU hot path is an
-MFMA-tiled GEMM on gfx950 (csrc/kernels/linear.hip): out = sigmoid(X @ W + b)
-over the encoded (one-hot + imputed-numeric) feature matrix.
+predict" as the tabular scoring hot path. The forest kernel covers RF and
+GBT traversal (creditcore.pack); dense linear scoring at scale lives in
+creditcore.dense (fused dense_score_kernel: impute + w·x + sigmoid per
+wavefront — the op is GEMV-shaped, so MFMA does not apply; batched
+multi-output models would route through rocBLAS GEMM instead).
 
-The CPU reference here is plain numpy fp64; training uses
-sklearn.linear_model.LogisticRegression on the same encoded matrix the forest
-pipeline produces.
+This module is the small CPU-side LinearScorer used for training-side
+experiments and as the numpy reference for the dense family's logistic
+head; training uses sklearn.linear_model.LogisticRegression on the encoded
+matrix the forest pipeline produces.
 """
 
 from __future__ import annotations
