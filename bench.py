@@ -127,10 +127,12 @@ def main():
     ]
     with_drift = not args.no_drift
 
-    if args.no_pipeline:
-
+    if args.no_pipeline or (args.rows <= 256 and device == "cuda"):
+        # latency mode: the fully-native single-call path (parse -> graph ->
+        # p-values -> response, one C++ call); pipelining buys nothing when
+        # the parse is microseconds.
         def one_step(i: int):
-            out = engine.score_json(pool[i % len(pool)])
+            out = engine.score_json_full(pool[i % len(pool)])
             assert out["rows"] == args.rows
             return out
 

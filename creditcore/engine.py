@@ -149,28 +149,61 @@ class ScoringEngine:
         from .pack import CATEGORICAL_FEATURES, MISSING_CATEGORY, NUMERIC_FEATURES
 
         if gpu.available():
-            enc = getattr(self, "_json_encoder", None)
-            if enc is None:
-                dc, dn = self.default_rows()
-                enc = gpu.ext().JsonEncoder(
-                    self.packed.vocabs,
-                    CATEGORICAL_FEATURES,
-                    NUMERIC_FEATURES,
-                    MISSING_CATEGORY,
-                    dc,
-                    dn,
-                )
-                self._json_encoder = enc
-            codes, nums = enc.encode(body)
+            codes, nums = self._ensure_json_encoder().encode(body)
             return np.asarray(codes), np.asarray(nums)
         import json
 
         return encode_batch(json.loads(body), self.packed.vocabs)
 
+    def _ensure_json_encoder(self):
+        enc = getattr(self, "_json_encoder", None)
+        if enc is None:
+            from .ops import gpu
+            from .pack import CATEGORICAL_FEATURES, MISSING_CATEGORY, NUMERIC_FEATURES
+
+            dc, dn = self.default_rows()
+            enc = gpu.ext().JsonEncoder(
+                self.packed.vocabs,
+                CATEGORICAL_FEATURES,
+                NUMERIC_FEATURES,
+                MISSING_CATEGORY,
+                dc,
+                dn,
+            )
+            self._json_encoder = enc
+        return enc
+
     def score_json(self, body: bytes) -> dict:
         """Score a raw /score JSON request body (wire-format fast path)."""
         codes, nums = self.encode_json_body(body)
         return self._score_encoded(codes, nums)
+
+    def score_json_full(self, body: bytes) -> dict:
+        """Fully-native request: one C++ call parses the wire JSON, stages,
+        replays the graph, converts drift p-values and serializes the
+        response — no Python object pass. Falls back to the staged path for
+        oversized batches or on CPU."""
+        if self.device != "cuda":
+            return self.score_json_bytes(body)
+        g = self._gpu
+        try:
+            resp, rows = g["sess"].score_json_full(
+                body,
+                self._ensure_json_encoder(),
+                self.packed.ref_cat_counts,
+                self.packed.ref_cat_offsets,
+                int(self.packed.ref_sorted_offsets[1] - self.packed.ref_sorted_offsets[0]),
+                FEATURES,
+            )
+        except RuntimeError:
+            # batch larger than the resident session capacity: grow + retry
+            # through the staged path
+            return self.score_json_bytes(body)
+        return {
+            "response_bytes": resp,
+            "rows": int(rows),
+            "cat_hist": g["np_hist"].copy(),
+        }
 
     def default_rows(self) -> tuple:
         """Encoded schema-default record (absent request fields take these

@@ -1284,27 +1284,23 @@ struct JsonEncoderState {
   }
 };
 
-py::tuple encode_json_impl(const JsonEncoderState& st, py::bytes body) {
+// Parse the full body into flat row-major code/num vectors. Caller must
+// NOT hold the GIL. Throws jsonenc::Error on malformed input.
+static size_t parse_records_nogil(const JsonEncoderState& st, char* data,
+                                  ssize_t blen, std::vector<int16_t>& codes,
+                                  std::vector<float>& nums) {
   const int ncat = st.ncat;
   const int nnum = st.nnum;
   const auto& vocab = st.vocab;
   const int16_t* def_codes = st.def_codes.data();
   const float* def_nums = st.def_nums.data();
-  char* data;
-  ssize_t blen;
-  if (PyBytes_AsStringAndSize(body.ptr(), &data, &blen) != 0)
-    throw py::value_error("body must be bytes");
-
-  std::vector<int16_t> codes;
-  std::vector<float> nums;
   size_t b = 0;
   {
-    py::gil_scoped_release nogil;
     jsonenc::Parser P{data, data + blen};
     P.begin_ = data;
     codes.reserve((size_t)blen / 70 * ncat + ncat);
     nums.reserve((size_t)blen / 70 * nnum + nnum);
-    try {
+    {
       P.expect('[', "body must be a JSON array");
       if (!P.eat(']')) {
         do {
@@ -1373,6 +1369,25 @@ py::tuple encode_json_impl(const JsonEncoderState& st, py::bytes body) {
       }
       P.ws();
       if (P.p != P.end) P.fail("trailing data");
+    }
+  }
+  return b;
+}
+
+py::tuple encode_json_impl(const JsonEncoderState& st, py::bytes body) {
+  const int ncat = st.ncat;
+  const int nnum = st.nnum;
+  char* data;
+  ssize_t blen;
+  if (PyBytes_AsStringAndSize(body.ptr(), &data, &blen) != 0)
+    throw py::value_error("body must be bytes");
+  std::vector<int16_t> codes;
+  std::vector<float> nums;
+  size_t b;
+  {
+    py::gil_scoped_release nogil;
+    try {
+      b = parse_records_nogil(st, data, blen, codes, nums);
     } catch (const jsonenc::Error& e) {
       py::gil_scoped_acquire gil;
       throw py::value_error(e.msg);
@@ -1478,19 +1493,9 @@ static double pelz_good_sf(double x, double n) {
   return std::min(1.0, std::max(0.0, 1.0 - cdf));
 }
 
-py::array_t<double> drift_pvals_host(
-    py::array_t<int32_t> batch_hist, py::array_t<float> ks_d,
-    py::array_t<int32_t> ref_cat_counts, py::array_t<int32_t> cat_offsets,
-    int64_t n_ref, int64_t n_batch) {
-  const auto* bh = batch_hist.data();
-  const auto* rc = ref_cat_counts.data();
-  const auto* off = cat_offsets.data();
-  const int ncat = (int)cat_offsets.size() - 1;
-  const auto* kd = ks_d.data();
-  const int nnum = (int)ks_d.size();
-
-  py::array_t<double> out({(ssize_t)(ncat + nnum)});
-  double* pv = out.mutable_data();
+static void drift_pvals_raw(const int32_t* bh, const float* kd, int nnum,
+                            const int32_t* rc, const int32_t* off, int ncat,
+                            int64_t n_ref, int64_t n_batch, double* pv) {
 
   for (int j = 0; j < ncat; ++j) {
     double rsum = 0, bsum = 0;
@@ -1519,6 +1524,17 @@ py::array_t<double> drift_pvals_host(
   const double en_f = (double)n_ref * (double)n_batch / ((double)n_ref + (double)n_batch);
   const double en = std::nearbyint(en_f);
   for (int j = 0; j < nnum; ++j) pv[ncat + j] = pelz_good_sf((double)kd[j], en);
+}
+
+py::array_t<double> drift_pvals_host(
+    py::array_t<int32_t> batch_hist, py::array_t<float> ks_d,
+    py::array_t<int32_t> ref_cat_counts, py::array_t<int32_t> cat_offsets,
+    int64_t n_ref, int64_t n_batch) {
+  const int ncat = (int)cat_offsets.size() - 1;
+  const int nnum = (int)ks_d.size();
+  py::array_t<double> out({(ssize_t)(ncat + nnum)});
+  drift_pvals_raw(batch_hist.data(), ks_d.data(), nnum, ref_cat_counts.data(),
+                  cat_offsets.data(), ncat, n_ref, n_batch, out.mutable_data());
   return out;
 }
 
@@ -1590,6 +1606,87 @@ py::bytes build_response_json(py::object pin_outs, int64_t b,
   return py::bytes(out);
 }
 
+// ---------------------------------------------------------------------------
+// The consolidated request path: one call = parse wire JSON -> pinned
+// staging -> graph replay -> drift p-values -> response JSON bytes. No
+// Python object pass at any stage (GIL released around parse/serialize;
+// ScoreSession::score releases it while waiting on the GPU).
+// ---------------------------------------------------------------------------
+
+py::tuple score_json_full(ScoreSession& s, py::bytes body,
+                          const JsonEncoderState& st,
+                          py::array_t<int32_t> ref_cat_counts,
+                          py::array_t<int32_t> cat_offsets, int64_t n_ref,
+                          py::list feature_names) {
+  char* data;
+  ssize_t blen;
+  if (PyBytes_AsStringAndSize(body.ptr(), &data, &blen) != 0)
+    throw py::value_error("body must be bytes");
+  const int nf = (int)py::len(feature_names);
+  TORCH_CHECK(nf == N_CAT + N_NUM, "feature name count mismatch");
+  std::vector<std::string> names(nf);
+  for (int j = 0; j < nf; ++j) names[j] = py::cast<std::string>(feature_names[j]);
+
+  std::vector<int16_t> codes;
+  std::vector<float> nums;
+  size_t b;
+  {
+    py::gil_scoped_release nogil;
+    try {
+      b = parse_records_nogil(st, data, blen, codes, nums);
+    } catch (const jsonenc::Error& e) {
+      py::gil_scoped_acquire gil;
+      throw py::value_error(e.msg);
+    }
+  }
+  if (b == 0) throw py::value_error("empty request batch");
+  TORCH_CHECK((int64_t)b <= s.capacity, "batch exceeds session capacity: ", b);
+
+  std::memcpy(s.pin_codes.data_ptr(), codes.data(), b * N_CAT * sizeof(int16_t));
+  std::memcpy(s.pin_nums.data_ptr(), nums.data(), b * N_NUM * sizeof(float));
+  const bool drift_now = (int64_t)b <= MAX_DRIFT_ROWS;
+  s.score((int64_t)b, drift_now, true);
+  int64_t nb = (int64_t)b;
+  if (!drift_now) {  // oversized: capped drift sample in a second pass
+    s.score(MAX_DRIFT_ROWS, true, true);
+    nb = MAX_DRIFT_ROWS;
+  }
+
+  double pv[N_CAT + N_NUM];
+  drift_pvals_raw(static_cast<const int32_t*>(s.pin_hist.data_ptr()),
+                  static_cast<const float*>(s.pin_ksd.data_ptr()), N_NUM,
+                  ref_cat_counts.data(), cat_offsets.data(),
+                  (int)cat_offsets.size() - 1, n_ref, nb, pv);
+
+  std::string out;
+  {
+    py::gil_scoped_release nogil;
+    const double* proba = s.pin_outs.data_ptr<double>();
+    const double* outlier = proba + 2 * b;
+    out.reserve(b * 24 + 2048);
+    out += "{\"predictions\": [";
+    for (size_t i = 0; i < b; ++i) {
+      if (i) out += ", ";
+      append_double(out, proba[i]);
+    }
+    out += "], \"outliers\": [";
+    for (size_t i = 0; i < b; ++i) {
+      if (i) out += ", ";
+      out += (outlier[i] != 0.0) ? "1.0" : "0.0";
+    }
+    out += "], \"feature_drift_batch\": {";
+    for (int j = 0; j < nf; ++j) {
+      if (j) out += ", ";
+      out += '\"';
+      out += names[j];
+      out += "\": ";
+      append_double(out, (double)(1.0f - (float)pv[j]));
+    }
+    out += "}}";
+  }
+  return py::make_tuple(py::bytes(out), (int64_t)b);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("score_forest_pipeline", &score_forest_pipeline,
         "Forest classifier + isolation forest scoring (gfx950)");
@@ -1625,6 +1722,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("score", &ScoreSession::score, py::arg("b"),
            py::arg("with_drift") = true, py::arg("sync") = true)
       .def("synchronize", &ScoreSession::synchronize)
+      .def("score_json_full", &score_json_full, py::arg("body"),
+           py::arg("encoder"), py::arg("ref_cat_counts"),
+           py::arg("cat_offsets"), py::arg("n_ref"), py::arg("feature_names"))
       .def_readonly("capacity", &ScoreSession::capacity)
       .def_property_readonly("pin_codes", [](ScoreSession& s) { return s.pin_codes; })
       .def_property_readonly("pin_nums", [](ScoreSession& s) { return s.pin_nums; })

@@ -138,6 +138,28 @@ def test_response_bytes_gpu_matches_dict_path(packed):
         assert abs(b["feature_drift_batch"][k] - v) < 1e-12
 
 
+@pytest.mark.gpu
+def test_score_json_full_matches_staged(packed):
+    import json
+
+    from creditcore.data import make_request_batch
+    from creditcore.engine import ScoringEngine
+
+    eng = ScoringEngine(packed, device="cuda")
+    for b in (1, 64, 1024):
+        body = json.dumps(make_request_batch(b, seed=b)).encode()
+        full = eng.score_json_full(body)
+        staged = eng.score_json_bytes(body)
+        assert full["rows"] == staged["rows"] == b
+        a = json.loads(full["response_bytes"])
+        c = json.loads(staged["response_bytes"])
+        assert a == c
+    with pytest.raises(ValueError):
+        eng.score_json_full(b"[]")
+    with pytest.raises(ValueError):
+        eng.score_json_full(b"not json")
+
+
 def test_serving_fallback_handles_lax_types(model_dir):
     """A numeric string coerces through the pydantic fallback exactly as the
     reference would (fast path rejects, fallback accepts)."""
