@@ -1,7 +1,7 @@
 """Scoring engine — one model replica on one device.
 
 The GPU path uploads the packed flat buffers (creditcore.pack) to HBM once and
-scores request batches with the HIP kernels (csrc/kernels/*.hip) on a private
+scores request batches with the HIP kernels (csrc/creditcore_kernels.hip) on a private
 HIP stream with pinned staging buffers:
 
     host:   encode strings -> codes (int16), nums (f32)      [request parse]

@@ -17,7 +17,7 @@ reference vocabulary are added as extra contingency columns (the natural
 generalisation — the reference never exercises unseen categories because its
 classifier uses handle_unknown="ignore").
 
-The GPU path (csrc/kernels/drift.hip) computes the K-S D statistic and the
+The GPU path (csrc/creditcore_kernels.hip (ks_kernel_t / cat_hist_kernel)) computes the K-S D statistic and the
 categorical histograms on-device; this module's ``chi2_from_counts`` /
 ``ks_asymp_pvalue`` convert those statistics to p-values identically on host,
 so CPU and GPU paths share the final numerics.

@@ -10,7 +10,7 @@ Reproduces the reference's sklearn pipeline exactly
 
 Training happens on CPU with sklearn (like the reference); *scoring* happens
 on MI355X via the packed flat-buffer representation (creditcore.pack) and the
-HIP forest-traversal kernel (csrc/kernels/score.hip). This module is the
+HIP forest-traversal kernel (csrc/creditcore_kernels.hip (forest_kernel_ilpN)). This module is the
 golden CPU path the GPU path is tested against.
 """
 

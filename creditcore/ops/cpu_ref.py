@@ -2,7 +2,7 @@
 
 This is (a) the golden reference the HIP kernels are tested against and
 (b) the no-GPU serving fallback. It deliberately uses float32 comparisons and
-arithmetic with the same semantics as csrc/kernels/score.hip, so GPU-vs-CPU
+arithmetic with the same semantics as csrc/creditcore_kernels.hip, so GPU-vs-CPU
 tests can use tight tolerances; CPU-ref-vs-sklearn tests use loose tolerances
 (f32 vs f64 threshold rounding can flip a measure-zero set of branches).
 """

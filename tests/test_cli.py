@@ -1,0 +1,48 @@
+"""CLI entrypoint coverage (python -m creditcore ...) — the reference's
+job/workflow invocations as subprocess smoke tests."""
+
+from __future__ import annotations
+
+import json
+import subprocess
+import sys
+
+
+def _run(*args, timeout=240):
+    return subprocess.run(
+        [sys.executable, "-m", "creditcore", *args],
+        capture_output=True,
+        text=True,
+        timeout=timeout,
+    )
+
+
+def test_usage_on_unknown_command():
+    r = _run("frobnicate")
+    assert r.returncode == 2
+    assert "usage" in r.stderr
+
+
+def test_train_pack_roundtrip(tmp_path):
+    model_dir = str(tmp_path / "model")
+    r = _run(
+        "train", "--model-dir", model_dir, "--max-evals", "1",
+        "--n-rows", "1200", "--no-register",
+    )
+    assert r.returncode == 0, r.stderr[-500:]
+    assert r.stdout.strip().endswith(model_dir)
+
+    r = _run("pack", "--model-dir", model_dir)
+    assert r.returncode == 0, r.stderr[-500:]
+    info = json.loads(r.stdout.strip().splitlines()[-1])
+    assert info["cls_trees"] >= 100
+    assert info["if_trees"] == 100
+    assert (tmp_path / "model" / "packed.npz").exists()
+
+
+def test_generate_data(tmp_path):
+    out = str(tmp_path / "c.csv")
+    r = _run("generate-data", "--out", out, "--n-rows", "500")
+    assert r.returncode == 0
+    head = open(out).readline()
+    assert head.startswith("sex,education,marriage")
