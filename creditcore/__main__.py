@@ -134,6 +134,30 @@ def _cmd_pipeline(argv):
     print(json.dumps(report, indent=2))
 
 
+def _cmd_train_dense(argv):
+    """Train the dense wide-tabular family (BASELINE config 5)."""
+    p = argparse.ArgumentParser(prog="creditcore train-dense")
+    p.add_argument("--model-dir", default="./dense_model")
+    p.add_argument("--train-rows", type=int, default=10_000_000)
+    p.add_argument("--feats", type=int, default=1000)
+    p.add_argument("--ref-rows", type=int, default=None)
+    p.add_argument("--epochs", type=int, default=1)
+    p.add_argument("--device", default="auto")
+    a = p.parse_args(argv)
+    from .dense import train_dense
+
+    model = train_dense(
+        n_rows=a.train_rows,
+        n_feats=a.feats,
+        ref_rows=a.ref_rows,
+        epochs=a.epochs,
+        device=a.device,
+    )
+    model.save(a.model_dir)
+    print(json.dumps({"model_dir": a.model_dir, "n_features": model.n_features,
+                      "ref_rows": model.n_ref}))
+
+
 def _cmd_generate_data(argv):
     """Write a UCI-shaped synthetic CSV (the reference's curated.csv analog,
     reference databricks/data/; the real UCI CSV is not available offline)."""
@@ -165,6 +189,7 @@ def main():
         "serve": _cmd_serve,
         "smoke": _cmd_smoke,
         "pipeline": _cmd_pipeline,
+        "train-dense": _cmd_train_dense,
         "generate-data": _cmd_generate_data,
     }
     if len(sys.argv) < 2 or sys.argv[1] not in cmds:

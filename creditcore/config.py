@@ -28,6 +28,8 @@ def _env(name: str, default, cast=None):
 class ServeConfig:
     # model + service (reference-compatible env names)
     model_directory: str = field(default_factory=lambda: _env("model_directory", "./model"))
+    # optional dense wide-tabular model dir (enables POST /predict_dense)
+    dense_model_dir: str = field(default_factory=lambda: _env("dense_model_dir", ""))
     service_name: str = field(default_factory=lambda: _env("service_name", "credit-default-api"))
     host: str = field(default_factory=lambda: _env("host", "0.0.0.0"))
     port: int = field(default_factory=lambda: _env("port", 5000, int))

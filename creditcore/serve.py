@@ -127,6 +127,14 @@ async def lifespan(app: FastAPI):
     ]
     for b in batchers:
         await b.start()
+    dense_engine = None
+    if cfg.dense_model_dir:
+        from .dense import DenseEngine, DenseModel
+
+        dense_engine = DenseEngine(
+            DenseModel.load(cfg.dense_model_dir), device=cfg.resolve_device()
+        )
+    state["dense_engine"] = dense_engine
     state["engines"] = engines
     state["batchers"] = batchers
     state["pool"] = ReplicaPool(len(engines))
@@ -280,6 +288,43 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
     @app.get("/metrics")
     async def metrics_endpoint():
         return state["metrics"].snapshot()
+
+    @app.post("/predict_dense")
+    async def predict_dense(request: Request):
+        """Dense wide-tabular scoring (BASELINE config 5). Body: binary
+        little-endian f32 — 8-byte header (uint32 rows, uint32 cols) then
+        rows*cols values. Enabled via --dense-model-dir."""
+        import struct
+
+        eng = state.get("dense_engine")
+        if eng is None:
+            raise HTTPException(status_code=404, detail="no dense model configured")
+        body = await request.body()
+        if len(body) < 8:
+            raise HTTPException(status_code=422, detail="missing rows/cols header")
+        rows, cols = struct.unpack("<II", body[:8])
+        if cols != eng.model.n_features:
+            raise HTTPException(
+                status_code=422,
+                detail=f"expected {eng.model.n_features} features, got {cols}",
+            )
+        expect = 8 + rows * cols * 4
+        if rows == 0 or len(body) != expect:
+            raise HTTPException(status_code=422, detail="body size mismatch")
+        x = np.frombuffer(body, dtype="<f4", offset=8).reshape(rows, cols)
+        t0 = time.perf_counter()
+        out = eng.score_arrays(x)
+        latency_ms = (time.perf_counter() - t0) * 1e3
+        state["metrics"].observe_request(rows, latency_ms)
+        return _json_response(
+            {
+                "predictions": np.asarray(out["predictions"]).tolist(),
+                "outliers": np.asarray(out["outliers"]).tolist(),
+                "feature_drift_batch": (
+                    np.float32(1.0) - np.asarray(out["p_vals"], dtype=np.float32)
+                ).astype(np.float64).tolist(),
+            }
+        )
 
     @app.get("/drift")
     async def drift_endpoint():

@@ -120,3 +120,44 @@ def test_dense_gpu_train_and_score():
     out = eng.score_arrays(x)
     assert np.isfinite(out["predictions"]).all()
     assert (out["p_vals"] > 1e-4).mean() > 0.9  # in-distribution
+
+
+def test_dense_serving_endpoint(dense_model, tmp_path):
+    """POST /predict_dense with a binary f32 body (BASELINE config 5 serve)."""
+    import struct
+
+    from fastapi.testclient import TestClient
+
+    from creditcore import train as T
+    from creditcore.config import ServeConfig
+    from creditcore.serve import create_app
+
+    dense_dir = str(tmp_path / "dense")
+    dense_model.save(dense_dir)
+    model_dir = str(tmp_path / "model")
+    T.train_and_register(model_dir=model_dir, max_evals=1, n_rows=1200, register=False)
+
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.dense_model_dir = dense_dir
+    cfg.device = "cpu"
+    with TestClient(create_app(cfg)) as client:
+        rng = np.random.default_rng(0)
+        x = rng.standard_normal((32, 64)).astype("<f4")
+        body = struct.pack("<II", 32, 64) + x.tobytes()
+        r = client.post("/predict_dense", content=body,
+                        headers={"content-type": "application/octet-stream"})
+        assert r.status_code == 200, r.text
+        out = r.json()
+        assert len(out["predictions"]) == 32
+        assert len(out["feature_drift_batch"]) == 64
+        # wrong width -> 422; no dense model -> 404 covered via fresh app
+        bad = struct.pack("<II", 1, 3) + b"\x00" * 12
+        assert client.post("/predict_dense", content=bad).status_code == 422
+
+    cfg2 = ServeConfig()
+    cfg2.model_directory = model_dir
+    cfg2.dense_model_dir = ""
+    cfg2.device = "cpu"
+    with TestClient(create_app(cfg2)) as client:
+        assert client.post("/predict_dense", content=b"").status_code == 404
