@@ -67,7 +67,14 @@ def main():
     p.add_argument("--no-drift", action="store_true")
     p.add_argument("--no-pipeline", action="store_true",
                    help="disable encode/GPU pipelining (pure closed loop)")
+    p.add_argument("--model-trees", type=int, default=None,
+                   help="override bench model n_estimators (sensitivity runs)")
+    p.add_argument("--model-depth", type=int, default=None)
     args = p.parse_args()
+    if args.model_trees:
+        BENCH_MODEL["n_estimators"] = args.model_trees
+    if args.model_depth:
+        BENCH_MODEL["max_depth"] = args.model_depth
 
     import torch
 
