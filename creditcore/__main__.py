@@ -158,6 +158,52 @@ def _cmd_train_dense(argv):
                       "ref_rows": model.n_ref}))
 
 
+def _cmd_score_batch(argv):
+    """Offline batch scoring job: CSV in, scored CSV out, sharded
+    round-robin across all visible GPUs (SURVEY.md §5.7 row-parallel
+    scaling; the reference's only offline path was pasting inference.csv
+    rows into Swagger)."""
+    p = argparse.ArgumentParser(prog="creditcore score-batch")
+    p.add_argument("--model-dir", default="./model")
+    p.add_argument("--input", required=True, help="CSV with the request schema")
+    p.add_argument("--output", required=True, help="CSV: input + prediction/outlier")
+    p.add_argument("--chunk-rows", type=int, default=16384)
+    p.add_argument("--device", default="auto")
+    a = p.parse_args(argv)
+    import numpy as np
+    import pandas as pd
+
+    from .config import ServeConfig
+    from .engine import ScoringEngine, load_engine
+    from .pack import encode_batch
+
+    device = a.device if a.device != "auto" else ServeConfig().resolve_device()
+    engines = [load_engine(a.model_dir, device=device, device_index=0)]
+    if device == "cuda":
+        import torch
+
+        for i in range(1, torch.cuda.device_count()):
+            engines.append(ScoringEngine(engines[0].packed, device="cuda", device_index=i))
+
+    df = pd.read_csv(a.input)
+    preds = np.empty(len(df))
+    outl = np.empty(len(df))
+    chunks = range(0, len(df), a.chunk_rows)
+    for k, lo in enumerate(chunks):
+        hi = min(lo + a.chunk_rows, len(df))
+        eng = engines[k % len(engines)]  # round-robin shard across GPUs
+        codes, nums = encode_batch(df.iloc[lo:hi], eng.packed.vocabs)
+        out = eng.score_arrays(codes, nums, with_drift=False)
+        preds[lo:hi] = out["predictions"]
+        outl[lo:hi] = out["outliers"]
+    df_out = df.copy()
+    df_out["prediction"] = preds
+    df_out["is_outlier"] = outl
+    df_out.to_csv(a.output, index=False)
+    print(json.dumps({"rows": len(df), "output": a.output,
+                      "engines": len(engines), "device": device}))
+
+
 def _cmd_generate_data(argv):
     """Write a UCI-shaped synthetic CSV (the reference's curated.csv analog,
     reference databricks/data/; the real UCI CSV is not available offline)."""
@@ -190,6 +236,7 @@ def main():
         "smoke": _cmd_smoke,
         "pipeline": _cmd_pipeline,
         "train-dense": _cmd_train_dense,
+        "score-batch": _cmd_score_batch,
         "generate-data": _cmd_generate_data,
     }
     if len(sys.argv) < 2 or sys.argv[1] not in cmds:

@@ -23,7 +23,11 @@ setup(
             sources=["csrc/creditcore_kernels.hip"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17"],
+                # CREDITCORE_ASAN=1: host-side AddressSanitizer debug build
+                # (SURVEY.md §5.2); run with LD_PRELOAD of libasan.
+                "nvcc": ["-O3", "-std=c++17"]
+                + (["-fsanitize=address", "-shared-libasan", "-g"]
+                   if os.environ.get("CREDITCORE_ASAN") == "1" else []),
             },
         )
     ],

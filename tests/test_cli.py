@@ -46,3 +46,21 @@ def test_generate_data(tmp_path):
     assert r.returncode == 0
     head = open(out).readline()
     assert head.startswith("sex,education,marriage")
+
+
+def test_score_batch(tmp_path):
+    model_dir = str(tmp_path / "model")
+    _run("train", "--model-dir", model_dir, "--max-evals", "1",
+         "--n-rows", "1200", "--no-register")
+    inp = str(tmp_path / "in.csv")
+    _run("generate-data", "--out", str(tmp_path / "_c.csv"),
+         "--n-rows", "10", "--inference-sample", inp)
+    out = str(tmp_path / "scored.csv")
+    r = _run("score-batch", "--model-dir", model_dir, "--input", inp,
+             "--output", out, "--device", "cpu", "--chunk-rows", "32")
+    assert r.returncode == 0, r.stderr[-500:]
+    import pandas as pd
+
+    df = pd.read_csv(out)
+    assert "prediction" in df and "is_outlier" in df and len(df) == 80
+    assert df["prediction"].between(0, 1).all()
