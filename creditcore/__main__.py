@@ -25,6 +25,9 @@ def _cmd_train(argv):
     p.add_argument("--n-rows", type=int, default=20_000)
     p.add_argument("--seed", type=int, default=2024)
     p.add_argument("--no-register", action="store_true")
+    p.add_argument("--min-roc-auc", type=float, default=None,
+                   help="refuse to register below this validation ROC-AUC "
+                        "(the quality gate the reference lacks)")
     p.add_argument("--algorithm", default="rf", choices=["rf", "gbt"],
                    help="rf = the reference's RandomForest; gbt = "
                         "gradient-boosted trees (same HIP traversal kernel)")
@@ -50,6 +53,7 @@ def _cmd_train(argv):
         register=not a.no_register,
         df=df,
         algorithm=a.algorithm,
+        min_roc_auc=a.min_roc_auc,
     )
     print(uri)
 

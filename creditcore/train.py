@@ -48,6 +48,10 @@ from .schema import CATEGORICAL_FEATURES, FEATURES, INPUT_SAMPLE, NUMERIC_FEATUR
 DEFAULT_MODEL_NAME = "credit-default-uci-custom"  # reference 02-register cell-13
 
 
+class QualityGateError(RuntimeError):
+    """Raised when the best run fails the registration quality gate."""
+
+
 @dataclass
 class EvalRun:
     run_id: str
@@ -159,6 +163,7 @@ def train_and_register(
     df: pd.DataFrame | None = None,
     register: bool = True,
     algorithm: str = "rf",
+    min_roc_auc: float | None = None,
 ) -> str:
     """The full train -> package -> register job (the reference's 2-task
     Databricks DAG, train_register_model.yml:10-39). Returns the model URI
@@ -166,6 +171,14 @@ def train_and_register(
     if df is None:
         df = make_uci_shaped_frame(n_rows=n_rows, seed=seed)
     best = train_model(df=df, max_evals=max_evals, seed=seed, algorithm=algorithm)
+    if min_roc_auc is not None:
+        auc = best.metrics["validation_roc_auc_score"]
+        if auc < min_roc_auc:
+            # the quality gate the reference lacks (metrics
+            # were logged but never enforced before registration/deploy)
+            raise QualityGateError(
+                f"best validation_roc_auc_score {auc:.4f} < gate {min_roc_auc}"
+            )
     drift, outlier = fit_detectors(df)
     registry.save_pyfunc_model(
         model_dir,

@@ -83,3 +83,28 @@ def test_run_artifacts_written(tmp_path):
     with open(os.path.join(runs, run_dirs[0], "run.json")) as f:
         rec = json.load(f)
     assert "params" in rec and "metrics" in rec
+
+
+def test_quality_gate_blocks_registration(tmp_path):
+    """--min-roc-auc refuses to package a model below the gate (the
+    enforcement the reference never had)."""
+    from creditcore.train import QualityGateError, train_and_register
+
+    df = make_uci_shaped_frame(n_rows=800, seed=1)
+    with pytest.raises(QualityGateError):
+        train_and_register(
+            model_dir=str(tmp_path / "m"),
+            max_evals=1,
+            df=df,
+            register=False,
+            min_roc_auc=0.999,  # unattainable
+        )
+    # a reachable gate passes
+    out = train_and_register(
+        model_dir=str(tmp_path / "m2"),
+        max_evals=1,
+        df=df,
+        register=False,
+        min_roc_auc=0.5,
+    )
+    assert out.endswith("m2")
