@@ -579,7 +579,8 @@ std::vector<torch::Tensor> drift_stats(
 
   int m_pow2 = 1;
   while (m_pow2 < B) m_pow2 <<= 1;
-  m_pow2 = std::max(m_pow2, 2);
+  // LDS must also hold one cross-wave reduction slot per wave
+  m_pow2 = std::max(m_pow2, BLOCK / 64);
   hipLaunchKernelGGL((ks_kernel_t<256>), dim3(N_NUM), dim3(BLOCK),
       (size_t)m_pow2 * sizeof(float), stream,
       nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, B, m_pow2,
@@ -746,7 +747,7 @@ torch::Tensor ks_stats(
   auto ks_d = torch::empty({F},
       torch::TensorOptions().dtype(torch::kFloat32).device(nums.device()));
   hipStream_t stream = c10::hip::getCurrentHIPStream();
-  int m_pow2 = 2;
+  int m_pow2 = (int)block / 64;  // >= one reduction slot per wave
   while (m_pow2 < B) m_pow2 <<= 1;
   size_t smem = (size_t)m_pow2 * sizeof(float);
   auto rs_off64 = rs_off.to(torch::kInt64);
@@ -917,7 +918,7 @@ struct ScoreSession {
           (size_t)total_bins * sizeof(int), stream2,
           d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
           hist.data_ptr<int>());
-      int m_pow2 = 2;
+      int m_pow2 = 512 / 64;  // >= one cross-wave reduction slot per wave
       while (m_pow2 < b) m_pow2 <<= 1;
       // Stage the reference column in LDS when batch + ref fit the 160 KiB
       // CU budget; searches then stay on-chip.
