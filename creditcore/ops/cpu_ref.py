@@ -27,31 +27,37 @@ def _traverse_forest(
     value_of: "callable",
     n_rows: int,
 ) -> np.ndarray:
-    """Sum of leaf values over all trees, per row (vectorised level-by-level)."""
+    """Sum of leaf values over all trees, per row.
+
+    Level-synchronous and vectorised over ALL (tree, row) pairs at once:
+    every pair advances one node per iteration, so the loop count is the
+    max tree depth (~20) instead of trees x depth — the per-tree version
+    cost ~8500 numpy dispatches for a single-row request."""
     bits = nodes[:, 1].view(np.float32)
+    T = len(offsets) - 1
+    starts = offsets[:-1].astype(np.int64)
+    # absolute node index per (tree, row) pair, flattened
+    cur = np.repeat(starts, n_rows)
+    base = cur.copy()
+    rows = np.tile(np.arange(n_rows, dtype=np.int64), T)
     acc = np.zeros(n_rows, dtype=np.float64)
-    rows_all = np.arange(n_rows)
-    for t in range(len(offsets) - 1):
-        cur = np.full(n_rows, offsets[t], dtype=np.int64)
-        active = rows_all.copy()
-        while len(active):
-            nidx = cur[active]
+    alive = np.arange(T * n_rows, dtype=np.int64)
+    while len(alive):
+        nidx = cur[alive]
+        feat = nodes[nidx, 0]
+        leaf = feat < 0
+        if leaf.any():
+            done = alive[leaf]
+            np.add.at(acc, rows[done], bits[cur[done]].astype(np.float64))
+            alive = alive[~leaf]
+            if not len(alive):
+                break
+            nidx = cur[alive]
             feat = nodes[nidx, 0]
-            leaf = feat < 0
-            if leaf.any():
-                lr = active[leaf]
-                acc[lr] += bits[cur[lr]].astype(np.float64)
-                active = active[~leaf]
-                if not len(active):
-                    break
-                nidx = cur[active]
-                feat = nodes[nidx, 0]
-            v = value_of(active, feat)
-            thr = bits[nidx]
-            go_left = v <= thr
-            nxt = np.where(go_left, nodes[nidx, 2], nodes[nidx, 3])
-            # children indices are per-tree-relative; make absolute
-            cur[active] = offsets[t] + nxt
+        v = value_of(rows[alive], feat)
+        go_left = v <= bits[nidx]
+        nxt = np.where(go_left, nodes[nidx, 2], nodes[nidx, 3])
+        cur[alive] = base[alive] + nxt  # children are tree-relative
     return acc
 
 

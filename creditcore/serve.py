@@ -43,7 +43,7 @@ def _build_engines(cfg: ServeConfig) -> list[ScoringEngine]:
         return [load_engine(cfg.model_directory, device="cpu")]
     import torch
 
-    n = cfg.n_gpus or torch.cuda.device_count()
+    n = min(cfg.n_gpus or torch.cuda.device_count(), torch.cuda.device_count())
     first = load_engine(cfg.model_directory, device="cuda", device_index=0)
     engines = [first]
     for i in range(1, n):
