@@ -960,9 +960,14 @@ struct ScoreSession {
     c10::hip::HIPGuard guard((c10::DeviceIndex)device_index);
 
     const uint64_t key = ((uint64_t)b << 1) | (with_drift ? 1 : 0);
+    // Only capture graphs for recurring shapes (powers of two and
+    // 256-multiples — what the micro-batcher and bench produce). Arbitrary
+    // merged sizes run the eager path directly: a capture costs ~ms, an
+    // eager pass costs ~30 µs extra — capture churn would swamp the win.
+    const bool graphable = ((b & (b - 1)) == 0) || (b % 256 == 0);
     auto it = graphs.find(key);
     if (it == graphs.end()) {
-      if (graphs.size() >= 64) {  // unbounded shapes: fall back to direct
+      if (!graphable || graphs.size() >= 64) {
         record(b, with_drift);
         if (sync) HIP_CHECK(hipStreamSynchronize(stream));
         return;
