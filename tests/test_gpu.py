@@ -94,6 +94,21 @@ def test_engine_refuses_silent_fallback(packed, monkeypatch):
         ScoringEngine(packed, device="cuda", device_index=0)
 
 
+def test_irregular_batch_sizes_eager_path(gpu_engine, packed):
+    """Non-bucketed sizes skip graph capture and run eagerly — results
+    must be identical to the CPU reference."""
+    rng = np.random.default_rng(0)
+    for b in (3, 97, 300, 777, 1500):
+        codes = np.stack(
+            [rng.integers(-1, len(v), size=b) for v in packed.vocabs], axis=1
+        ).astype(np.int16)
+        nums = rng.normal(1000.0, 500.0, size=(b, 14)).astype(np.float32)
+        out = gpu_engine.score_arrays(codes, nums)
+        ref = cpu_ref.score_batch_cpu(packed, codes, nums)
+        np.testing.assert_allclose(out["predictions"], ref["predictions"], atol=1e-9)
+        np.testing.assert_allclose(out["p_vals"], ref["p_vals"], atol=1e-6)
+
+
 def test_serving_stack_on_gpu(model_dir):
     """Whole serving stack on the GPU engine (TestClient): contract-valid
     responses through both the solo-flush bytes path and the merged path."""
