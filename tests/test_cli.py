@@ -64,3 +64,15 @@ def test_score_batch(tmp_path):
     df = pd.read_csv(out)
     assert "prediction" in df and "is_outlier" in df and len(df) == 80
     assert df["prediction"].between(0, 1).all()
+
+
+def test_config_env_and_cli_precedence(monkeypatch):
+    from creditcore.config import ServeConfig
+
+    monkeypatch.setenv("CREDITCORE_PORT", "5055")
+    monkeypatch.setenv("MODEL_DIRECTORY", "/env/model")  # reference env name
+    cfg = ServeConfig()
+    assert cfg.port == 5055
+    assert cfg.model_directory == "/env/model"
+    cfg2 = ServeConfig.from_args(["--port", "6000"])
+    assert cfg2.port == 6000  # CLI wins over env default

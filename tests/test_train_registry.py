@@ -108,3 +108,17 @@ def test_quality_gate_blocks_registration(tmp_path):
         min_roc_auc=0.5,
     )
     assert out.endswith("m2")
+
+
+def test_resolve_unknown_model_raises(tmp_path):
+    with pytest.raises(FileNotFoundError):
+        registry.resolve_model_uri("models:/nope/latest", str(tmp_path))
+
+
+def test_resolve_plain_path_passthrough():
+    assert registry.resolve_model_uri("/some/dir") == "/some/dir"
+
+
+def test_load_pyfunc_rejects_non_model_dir(tmp_path):
+    with pytest.raises(FileNotFoundError):
+        registry.load_pyfunc_model(str(tmp_path))
