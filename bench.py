@@ -166,6 +166,27 @@ def main():
         executor = ThreadPoolExecutor(max_workers=2)
         DEPTH = 3  # encode i+1..i+3 overlap scoring of i (C parser drops the GIL)
 
+        use_slots = device == "cuda" and args.rows <= 16384
+
+        def _post_step(out, nums, i, outs, step_times, t_prev):
+            assert out["rows"] == args.rows
+            if (
+                drift_sync is not None
+                and "cat_hist" in out
+                and (i + 1) % DRIFT_SYNC_PERIOD == 0
+            ):
+                drift_sync.accumulate(
+                    torch.from_numpy(out["cat_hist"]).to(drift_dev),
+                    torch.from_numpy(nums).to(drift_dev),
+                )
+                drift_sync.allreduce()
+            outs.append(out)
+            if step_times is not None:
+                t_now = time.perf_counter()
+                step_times.append(t_now - t_prev)
+                return t_now
+            return t_prev
+
         def run_steps(k: int, step_times=None):
             outs = []
             q = deque(
@@ -173,6 +194,7 @@ def main():
                 for i in range(min(DEPTH, k))
             )
             t_prev = time.perf_counter()
+            pending = None  # (slot, rows, nums) awaiting epilogue
             for i in range(k):
                 codes, nums = q.popleft().result()
                 nxt = i + DEPTH
@@ -180,23 +202,22 @@ def main():
                     q.append(
                         executor.submit(engine.encode_json_body, pool[nxt % len(pool)])
                     )
-                out = engine.score_encoded_bytes(codes, nums)
-                assert out["rows"] == args.rows
-                if (
-                    drift_sync is not None
-                    and "cat_hist" in out
-                    and (i + 1) % DRIFT_SYNC_PERIOD == 0
-                ):
-                    drift_sync.accumulate(
-                        torch.from_numpy(out["cat_hist"]).to(drift_dev),
-                        torch.from_numpy(nums).to(drift_dev),
-                    )
-                    drift_sync.allreduce()
-                outs.append(out)
-                if step_times is not None:
-                    t_now = time.perf_counter()
-                    step_times.append(t_now - t_prev)
-                    t_prev = t_now
+                if use_slots:
+                    # launch step i's graph async, then run step i-1's
+                    # epilogue (p-values + response serialization) while the
+                    # GPU executes step i — double-buffered pinned slots
+                    slot = i & 1
+                    b = engine.submit_encoded_slot(codes, nums, slot)
+                    if pending is not None:
+                        out = engine.finish_slot(pending[0], pending[1])
+                        t_prev = _post_step(out, pending[2], len(outs), outs, step_times, t_prev)
+                    pending = (slot, b, nums)
+                else:
+                    out = engine.score_encoded_bytes(codes, nums)
+                    t_prev = _post_step(out, nums, len(outs), outs, step_times, t_prev)
+            if pending is not None:
+                out = engine.finish_slot(pending[0], pending[1])
+                _post_step(out, pending[2], len(outs), outs, step_times, t_prev)
             return outs
 
     run_steps(args.warmup)

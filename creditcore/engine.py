@@ -71,29 +71,25 @@ class ScoringEngine:
             "if_threshold": float(p.if_threshold),
         }
         sess = ext.ScoreSession(model, capacity, self.device_index)
-        self._gpu = {
-            "ext": ext,
-            "model": model,
-            "sess": sess,
-            # zero-copy numpy views of the session's pinned staging/outputs
-            "np_codes": sess.pin_codes.numpy(),
-            "np_nums": sess.pin_nums.numpy(),
-            "np_outs": sess.pin_outs.numpy(),
-            "np_hist": sess.pin_hist.numpy(),
-            "np_ksd": sess.pin_ksd.numpy(),
-        }
+        self._gpu = {"ext": ext, "model": model}
+        self._set_session(sess)
+
+    def _set_session(self, sess):
+        g = self._gpu
+        g["sess"] = sess
+        # zero-copy numpy views of the session's pinned staging/outputs;
+        # first index = slot (double-buffered for pipelined callers)
+        g["np_codes"] = sess.pin_codes.numpy()
+        g["np_nums"] = sess.pin_nums.numpy()
+        g["np_outs"] = sess.pin_outs.numpy()
+        g["np_hist"] = sess.pin_hist.numpy()
+        g["np_ksd"] = sess.pin_ksd.numpy()
 
     def _ensure_capacity(self, b: int):
         g = self._gpu
         if b > g["sess"].capacity:
             cap = 1 << (b - 1).bit_length()
-            sess = g["ext"].ScoreSession(g["model"], cap, self.device_index)
-            g["sess"] = sess
-            g["np_codes"] = sess.pin_codes.numpy()
-            g["np_nums"] = sess.pin_nums.numpy()
-            g["np_outs"] = sess.pin_outs.numpy()
-            g["np_hist"] = sess.pin_hist.numpy()
-            g["np_ksd"] = sess.pin_ksd.numpy()
+            self._set_session(g["ext"].ScoreSession(g["model"], cap, self.device_index))
 
     def _score_gpu(self, codes: np.ndarray, nums: np.ndarray, with_drift: bool = True) -> dict:
         g = self._gpu
@@ -102,10 +98,10 @@ class ScoringEngine:
         # Drift is a batch-population statistic; cap its sample at the K-S
         # kernel's LDS sort capacity (predictions still cover every row).
         drift_now = with_drift and b <= self.DRIFT_MAX_ROWS
-        g["np_codes"][:b] = codes
-        g["np_nums"][:b] = nums
+        g["np_codes"][0][:b] = codes
+        g["np_nums"][0][:b] = nums
         g["sess"].score(b, drift_now, True)  # blocks; GIL released
-        flat = g["np_outs"].reshape(-1)  # b-packed: proba | iscore | outlier
+        flat = g["np_outs"][0].reshape(-1)  # b-packed: proba | iscore | outlier
         out = {
             "predictions": flat[:b].copy(),
             "instance_score": flat[b : 2 * b].copy(),
@@ -117,8 +113,8 @@ class ScoringEngine:
             g["sess"].score(db, True, True)
             drift_now, b = True, db
         if drift_now:
-            hist = g["np_hist"].copy()
-            ks_d = g["np_ksd"].copy()
+            hist = g["np_hist"][0].copy()
+            ks_d = g["np_ksd"][0].copy()
             out["p_vals"] = cpu_ref.pvals_from_stats(self.packed, hist, ks_d, b)
             out["cat_hist"] = hist
             out["ks_d"] = ks_d
@@ -202,7 +198,7 @@ class ScoringEngine:
         return {
             "response_bytes": resp,
             "rows": int(rows),
-            "cat_hist": g["np_hist"].copy(),
+            "cat_hist": g["np_hist"][0].copy(),
         }
 
     def default_rows(self) -> tuple:
@@ -239,14 +235,14 @@ class ScoringEngine:
         g = self._gpu
         self._ensure_capacity(b)
         drift_now = b <= self.DRIFT_MAX_ROWS
-        g["np_codes"][:b] = codes
-        g["np_nums"][:b] = nums
+        g["np_codes"][0][:b] = codes
+        g["np_nums"][0][:b] = nums
         g["sess"].score(b, drift_now, True)
         nb = b
         if not drift_now:
             g["sess"].score(self.DRIFT_MAX_ROWS, True, True)
             nb = self.DRIFT_MAX_ROWS
-        pvals = cpu_ref.pvals_from_stats(self.packed, g["np_hist"], g["np_ksd"], nb)
+        pvals = cpu_ref.pvals_from_stats(self.packed, g["np_hist"][0], g["np_ksd"][0], nb)
         resp = g["ext"].build_response_json(
             g["sess"].pin_outs, b, np.ascontiguousarray(pvals), FEATURES
         )
@@ -254,7 +250,38 @@ class ScoringEngine:
             "response_bytes": resp,
             "rows": b,
             # for node-global drift aggregation by the serving layer
-            "cat_hist": g["np_hist"].copy(),
+            "cat_hist": g["np_hist"][0].copy(),
+        }
+
+    # -------------------------------------------------- pipelined slot API
+    def submit_encoded_slot(self, codes: np.ndarray, nums: np.ndarray, slot: int) -> int:
+        """Stage + launch (no sync) into a pinned slot; pair with
+        finish_slot. Slot i%2 lets step i's epilogue overlap step i+1's
+        graph. Falls back to capacity growth like the sync path."""
+        g = self._gpu
+        b = len(codes)
+        self._ensure_capacity(b)
+        assert b <= self.DRIFT_MAX_ROWS, "pipelined path caps at DRIFT_MAX_ROWS"
+        s = slot & 1
+        g["np_codes"][s][:b] = codes
+        g["np_nums"][s][:b] = nums
+        g["sess"].score(b, True, False, s)
+        return b
+
+    def finish_slot(self, slot: int, b: int) -> dict:
+        """Wait for the slot's graph; C epilogue: drift p-values +
+        response serialization from the slot's pinned buffers."""
+        g = self._gpu
+        s = slot & 1
+        n_ref = int(self.packed.ref_sorted_offsets[1] - self.packed.ref_sorted_offsets[0])
+        resp = g["sess"].response_epilogue(
+            s, b, self.packed.ref_cat_counts, self.packed.ref_cat_offsets,
+            n_ref, FEATURES,
+        )
+        return {
+            "response_bytes": resp,
+            "rows": b,
+            "cat_hist": g["np_hist"][s].copy(),
         }
 
     def score_records(self, records) -> dict:

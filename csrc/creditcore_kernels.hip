@@ -795,6 +795,14 @@ struct ScoreSession {
   hipStream_t stream{};
   hipStream_t stream2{};  // drift branch (runs parallel to the forests)
   hipEvent_t ev_fork{}, ev_join{};
+  hipEvent_t ev_done[2]{};  // per-slot completion (async score)
+
+  // raw slot pointers into the pinned buffers
+  int16_t* p_codes(int s) { return pin_codes.data_ptr<int16_t>() + (size_t)s * capacity * N_CAT; }
+  float* p_nums(int s) { return pin_nums.data_ptr<float>() + (size_t)s * capacity * N_NUM; }
+  double* p_outs(int s) { return pin_outs.data_ptr<double>() + (size_t)s * 3 * capacity; }
+  int32_t* p_hist(int s) { return pin_hist.data_ptr<int32_t>() + (size_t)s * total_bins; }
+  float* p_ksd(int s) { return pin_ksd.data_ptr<float>() + (size_t)s * N_NUM; }
 
   ScoreSession(py::dict model, int64_t cap, int dev) : capacity(cap), device_index(dev) {
     c10::hip::HIPGuard guard((c10::DeviceIndex)dev);
@@ -832,12 +840,14 @@ struct ScoreSession {
     hist = torch::empty({total_bins}, devopt.dtype(torch::kInt32));
     ksd = torch::empty({N_NUM}, devopt.dtype(torch::kFloat32));
 
+    // two slots: step i's host epilogue reads slot i%2 while step i+1's
+    // graph fills the other slot (pipelined serving/bench loops)
     auto pinned = torch::TensorOptions().pinned_memory(true);
-    pin_codes = torch::empty({capacity, N_CAT}, pinned.dtype(torch::kInt16));
-    pin_nums = torch::empty({capacity, N_NUM}, pinned.dtype(torch::kFloat32));
-    pin_outs = torch::empty({3, capacity}, pinned.dtype(torch::kFloat64));
-    pin_hist = torch::empty({total_bins}, pinned.dtype(torch::kInt32));
-    pin_ksd = torch::empty({N_NUM}, pinned.dtype(torch::kFloat32));
+    pin_codes = torch::empty({2, capacity, N_CAT}, pinned.dtype(torch::kInt16));
+    pin_nums = torch::empty({2, capacity, N_NUM}, pinned.dtype(torch::kFloat32));
+    pin_outs = torch::empty({2, 3, capacity}, pinned.dtype(torch::kFloat64));
+    pin_hist = torch::empty({2, total_bins}, pinned.dtype(torch::kInt32));
+    pin_ksd = torch::empty({2, N_NUM}, pinned.dtype(torch::kFloat32));
 
     auto rs = py::cast<torch::Tensor>(model["ref_sorted_offsets"]);
     auto rs_acc = rs.accessor<int32_t, 1>();
@@ -852,6 +862,8 @@ struct ScoreSession {
     HIP_CHECK(hipStreamCreateWithFlags(&stream2, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_join, hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&ev_done[0], hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&ev_done[1], hipEventDisableTiming));
     HIP_CHECK(hipDeviceSynchronize());  // uploads above used torch's stream
   }
 
@@ -863,16 +875,18 @@ struct ScoreSession {
       (void)hipStreamDestroy(stream2);
       (void)hipEventDestroy(ev_fork);
       (void)hipEventDestroy(ev_join);
+      (void)hipEventDestroy(ev_done[0]);
+      (void)hipEventDestroy(ev_done[1]);
     }
   }
 
   // Record the full scoring sequence for batch size b on `stream`.
   // Output layout (b-packed so one D2H covers all three): outs holds
   // proba[0:b] | iscore[b:2b] | outlier[2b:3b]; pin_outs mirrors it.
-  void record(int b, bool with_drift) {
-    HIP_CHECK(hipMemcpyAsync(d_codes.data_ptr(), pin_codes.data_ptr(),
+  void record(int b, bool with_drift, int slot) {
+    HIP_CHECK(hipMemcpyAsync(d_codes.data_ptr(), p_codes(slot),
         (size_t)b * N_CAT * sizeof(short), hipMemcpyHostToDevice, stream));
-    HIP_CHECK(hipMemcpyAsync(d_nums.data_ptr(), pin_nums.data_ptr(),
+    HIP_CHECK(hipMemcpyAsync(d_nums.data_ptr(), p_nums(slot),
         (size_t)b * N_NUM * sizeof(float), hipMemcpyHostToDevice, stream));
     // fork point: the drift branch (stream2) depends only on the H2D copies
     if (with_drift) {
@@ -936,14 +950,14 @@ struct ScoreSession {
           d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b, m_pow2,
           (int)ref_lds, ref_sorted.data_ptr<float>(),
           rs_off.data_ptr<int64_t>(), ksd.data_ptr<float>());
-      HIP_CHECK(hipMemcpyAsync(pin_hist.data_ptr(), hist.data_ptr(),
+      HIP_CHECK(hipMemcpyAsync(p_hist(slot), hist.data_ptr(),
           (size_t)total_bins * sizeof(int), hipMemcpyDeviceToHost, stream2));
-      HIP_CHECK(hipMemcpyAsync(pin_ksd.data_ptr(), ksd.data_ptr(),
+      HIP_CHECK(hipMemcpyAsync(p_ksd(slot), ksd.data_ptr(),
           (size_t)N_NUM * sizeof(float), hipMemcpyDeviceToHost, stream2));
       HIP_CHECK(hipEventRecord(ev_join, stream2));
       HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
     }
-    HIP_CHECK(hipMemcpyAsync(pin_outs.data_ptr(), proba,
+    HIP_CHECK(hipMemcpyAsync(p_outs(slot), proba,
         (size_t)(3 * b) * sizeof(double), hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipGetLastError());
   }
@@ -952,14 +966,16 @@ struct ScoreSession {
   // released) until pin_outs/pin_hist/pin_ksd hold the results. The whole
   // sequence is captured into a hipGraph per (b, with_drift) and replayed
   // as one submit on subsequent requests of the same shape.
-  void score(int64_t b64, bool with_drift, bool sync) {
+  void score(int64_t b64, bool with_drift, bool sync, int64_t slot64) {
     TORCH_CHECK(b64 >= 1 && b64 <= capacity, "batch out of range: ", b64);
     TORCH_CHECK(!with_drift || b64 <= MAX_DRIFT_ROWS, "drift batch too large: ", b64);
+    TORCH_CHECK(slot64 == 0 || slot64 == 1, "slot must be 0/1");
     const int b = (int)b64;
+    const int slot = (int)slot64;
     py::gil_scoped_release nogil;
     c10::hip::HIPGuard guard((c10::DeviceIndex)device_index);
 
-    const uint64_t key = ((uint64_t)b << 1) | (with_drift ? 1 : 0);
+    const uint64_t key = ((uint64_t)b << 2) | ((with_drift ? 1u : 0u) << 1) | (uint64_t)slot;
     // Only capture graphs for recurring shapes (powers of two and
     // 256-multiples — what the micro-batcher and bench produce). Arbitrary
     // merged sizes run the eager path directly: a capture costs ~ms, an
@@ -967,14 +983,15 @@ struct ScoreSession {
     const bool graphable = ((b & (b - 1)) == 0) || (b % 256 == 0);
     auto it = graphs.find(key);
     if (it == graphs.end()) {
-      if (!graphable || graphs.size() >= 64) {
-        record(b, with_drift);
+      if (!graphable || graphs.size() >= 128) {
+        record(b, with_drift, slot);
         if (sync) HIP_CHECK(hipStreamSynchronize(stream));
+        else HIP_CHECK(hipEventRecord(ev_done[slot], stream));
         return;
       }
       hipGraph_t graph;
       HIP_CHECK(hipStreamBeginCapture(stream, hipStreamCaptureModeThreadLocal));
-      record(b, with_drift);
+      record(b, with_drift, slot);
       HIP_CHECK(hipStreamEndCapture(stream, &graph));
       hipGraphExec_t exec;
       HIP_CHECK(hipGraphInstantiate(&exec, graph, nullptr, nullptr, 0));
@@ -983,6 +1000,12 @@ struct ScoreSession {
     }
     HIP_CHECK(hipGraphLaunch(it->second, stream));
     if (sync) HIP_CHECK(hipStreamSynchronize(stream));
+    else HIP_CHECK(hipEventRecord(ev_done[slot], stream));
+  }
+
+  void wait_slot(int64_t slot) {
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipEventSynchronize(ev_done[slot & 1]));
   }
 
   std::unordered_map<uint64_t, hipGraphExec_t> graphs;
@@ -1724,26 +1747,25 @@ py::tuple score_json_full(ScoreSession& s, py::bytes body,
   if (b == 0) throw py::value_error("empty request batch");
   TORCH_CHECK((int64_t)b <= s.capacity, "batch exceeds session capacity: ", b);
 
-  std::memcpy(s.pin_codes.data_ptr(), codes.data(), b * N_CAT * sizeof(int16_t));
-  std::memcpy(s.pin_nums.data_ptr(), nums.data(), b * N_NUM * sizeof(float));
+  std::memcpy(s.p_codes(0), codes.data(), b * N_CAT * sizeof(int16_t));
+  std::memcpy(s.p_nums(0), nums.data(), b * N_NUM * sizeof(float));
   const bool drift_now = (int64_t)b <= MAX_DRIFT_ROWS;
-  s.score((int64_t)b, drift_now, true);
+  s.score((int64_t)b, drift_now, true, 0);
   int64_t nb = (int64_t)b;
   if (!drift_now) {  // oversized: capped drift sample in a second pass
-    s.score(MAX_DRIFT_ROWS, true, true);
+    s.score(MAX_DRIFT_ROWS, true, true, 0);
     nb = MAX_DRIFT_ROWS;
   }
 
   double pv[N_CAT + N_NUM];
-  drift_pvals_raw(static_cast<const int32_t*>(s.pin_hist.data_ptr()),
-                  static_cast<const float*>(s.pin_ksd.data_ptr()), N_NUM,
+  drift_pvals_raw(s.p_hist(0), s.p_ksd(0), N_NUM,
                   ref_cat_counts.data(), cat_offsets.data(),
                   (int)cat_offsets.size() - 1, n_ref, nb, pv);
 
   std::string out;
   {
     py::gil_scoped_release nogil;
-    const double* proba = s.pin_outs.data_ptr<double>();
+    const double* proba = s.p_outs(0);
     const double* outlier = proba + 2 * b;
     out.reserve(b * 24 + 2048);
     out += "{\"predictions\": [";
@@ -1767,6 +1789,54 @@ py::tuple score_json_full(ScoreSession& s, py::bytes body,
     out += "}}";
   }
   return py::make_tuple(py::bytes(out), (int64_t)b);
+}
+
+// Pipelined epilogue: wait for slot's graph, convert drift p-values and
+// serialize the response from that slot's pinned buffers.
+py::bytes response_epilogue(ScoreSession& s, int64_t slot64, int64_t b64,
+                            py::array_t<int32_t> ref_cat_counts,
+                            py::array_t<int32_t> cat_offsets, int64_t n_ref,
+                            py::list feature_names) {
+  const int slot = (int)slot64 & 1;
+  const int64_t b = b64;
+  TORCH_CHECK(b >= 1 && b <= s.capacity, "batch out of range: ", b);
+  const int nf = (int)py::len(feature_names);
+  std::vector<std::string> names(nf);
+  for (int j = 0; j < nf; ++j) names[j] = py::cast<std::string>(feature_names[j]);
+  const int64_t nb = std::min<int64_t>(b, MAX_DRIFT_ROWS);
+
+  double pv[N_CAT + N_NUM];
+  std::string out;
+  {
+    py::gil_scoped_release nogil;
+    HIP_CHECK(hipEventSynchronize(s.ev_done[slot]));
+    drift_pvals_raw(s.p_hist(slot), s.p_ksd(slot), N_NUM,
+                    ref_cat_counts.data(), cat_offsets.data(),
+                    (int)cat_offsets.size() - 1, n_ref, nb, pv);
+    const double* proba = s.p_outs(slot);
+    const double* outlier = proba + 2 * b;
+    out.reserve((size_t)b * 24 + 2048);
+    out += "{\"predictions\": [";
+    for (int64_t i = 0; i < b; ++i) {
+      if (i) out += ", ";
+      append_double(out, proba[i]);
+    }
+    out += "], \"outliers\": [";
+    for (int64_t i = 0; i < b; ++i) {
+      if (i) out += ", ";
+      out += (outlier[i] != 0.0) ? "1.0" : "0.0";
+    }
+    out += "], \"feature_drift_batch\": {";
+    for (int j = 0; j < nf; ++j) {
+      if (j) out += ", ";
+      out += '\"';
+      out += names[j];
+      out += "\": ";
+      append_double(out, (double)(1.0f - (float)pv[j]));
+    }
+    out += "}}";
+  }
+  return py::bytes(out);
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -1802,7 +1872,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def(py::init<py::dict, int64_t, int>(), py::arg("model"),
            py::arg("capacity"), py::arg("device_index"))
       .def("score", &ScoreSession::score, py::arg("b"),
-           py::arg("with_drift") = true, py::arg("sync") = true)
+           py::arg("with_drift") = true, py::arg("sync") = true,
+           py::arg("slot") = 0)
+      .def("wait_slot", &ScoreSession::wait_slot)
+      .def("response_epilogue", &response_epilogue, py::arg("slot"),
+           py::arg("b"), py::arg("ref_cat_counts"), py::arg("cat_offsets"),
+           py::arg("n_ref"), py::arg("feature_names"))
       .def("synchronize", &ScoreSession::synchronize)
       .def("score_json_full", &score_json_full, py::arg("body"),
            py::arg("encoder"), py::arg("ref_cat_counts"),

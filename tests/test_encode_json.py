@@ -160,6 +160,35 @@ def test_score_json_full_matches_staged(packed):
         eng.score_json_full(b"not json")
 
 
+@pytest.mark.gpu
+def test_slot_pipeline_matches_sync(packed):
+    """submit_encoded_slot/finish_slot (double-buffered async path) must
+    produce byte-identical responses to the synchronous path, including
+    when slots alternate across differing batches."""
+    import json as _json
+
+    from creditcore.data import make_request_batch
+    from creditcore.engine import ScoringEngine
+    from creditcore.pack import encode_batch
+
+    eng = ScoringEngine(packed, device="cuda")
+    batches = [
+        encode_batch(make_request_batch(b, seed=40 + b), packed.vocabs)
+        for b in (64, 257, 64, 1024, 513)
+    ]
+    want = [eng.score_encoded_bytes(c, n)["response_bytes"] for c, n in batches]
+    got = [None] * len(batches)
+    pending = None
+    for i, (c, n) in enumerate(batches):
+        b = eng.submit_encoded_slot(c, n, i & 1)
+        if pending is not None:
+            got[i - 1] = eng.finish_slot(pending[0], pending[1])["response_bytes"]
+        pending = (i & 1, b)
+    got[-1] = eng.finish_slot(pending[0], pending[1])["response_bytes"]
+    for w, g in zip(want, got):
+        assert _json.loads(w) == _json.loads(g)
+
+
 def test_serving_fallback_handles_lax_types(model_dir):
     """A numeric string coerces through the pydantic fallback exactly as the
     reference would (fast path rejects, fallback accepts)."""
