@@ -19,6 +19,7 @@ from __future__ import annotations
 
 import asyncio
 import time
+from collections import deque
 from dataclasses import dataclass, field
 
 import numpy as np
@@ -43,7 +44,7 @@ class MicroBatcher:
         self.score_single = score_single
         self.max_rows = int(max_rows)
         self.max_wait = max(0.0, max_wait_us * 1e-6)
-        self._pending: list[_Pending] = []
+        self._pending: deque[_Pending] = deque()
         self._pending_rows = 0
         self._event = asyncio.Event()
         self._task: asyncio.Task | None = None
@@ -88,15 +89,15 @@ class MicroBatcher:
             await self._flush()
 
     async def _flush(self):
-        pending, self._pending = self._pending, []
+        pending, self._pending = self._pending, deque()
         self._pending_rows = 0
         # max_rows is a hard cap per engine call: chunk greedily by request
         # boundaries (one oversized request still goes through whole).
         while pending:
-            batch = [pending.pop(0)]
+            batch = [pending.popleft()]
             rows = len(batch[0].codes)
             while pending and rows + len(pending[0].codes) <= self.max_rows:
-                p = pending.pop(0)
+                p = pending.popleft()
                 batch.append(p)
                 rows += len(p.codes)
             await self._flush_one(batch)
