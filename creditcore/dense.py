@@ -200,6 +200,14 @@ class DenseEngine:
     MAX_DRIFT_ROWS = 65536  # sort+scan path; the in-LDS kernel caps at 16384
 
     def __init__(self, model: DenseModel, device: str = "cpu", device_index: int = 0):
+        if device == "cpu":
+            try:
+                import torch
+
+                if torch.is_tensor(model.weight):
+                    model = model.as_numpy()  # CPU path computes in numpy
+            except ImportError:  # pragma: no cover
+                pass
         self.model = model
         self.device = device
         self.device_index = device_index
