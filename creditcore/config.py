@@ -34,6 +34,7 @@ class ServeConfig:
     host: str = field(default_factory=lambda: _env("host", "0.0.0.0"))
     port: int = field(default_factory=lambda: _env("port", 5000, int))
     workers: int = field(default_factory=lambda: _env("workers", 1, int))
+    raw_http: bool = field(default_factory=lambda: _env("raw_http", False, bool))
 
     # device
     device: str = field(default_factory=lambda: _env("device", "auto"))  # auto|cuda|cpu

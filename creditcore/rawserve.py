@@ -1,0 +1,241 @@
+"""Raw asyncio HTTP/1.1 frontend — the serving-layer fast path.
+
+uvicorn in this stack parses requests with pure-Python h11, which caps the
+wire layer far below the engine (~0.65 MB request bodies). This frontend
+speaks just enough HTTP/1.1 for the scoring contract — request line,
+headers, Content-Length bodies, keep-alive — and hands the body bytes
+straight to the native JSON fast path. The FastAPI app (`creditcore.serve`)
+remains the contract-complete default (docs, OpenAPI, pydantic fallback
+validation); enable this one with `creditcore serve --raw-http` where
+throughput matters.
+
+Endpoints: POST /score, /predict (wire-format fast path with pydantic
+fallback semantics preserved via the same `_encode` logic), GET /healthz,
+/metrics, /drift.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+import uuid
+
+import numpy as np
+
+from .batching import MicroBatcher
+from .config import ServeConfig
+from .schema import FEATURES, LoanApplicant
+from .utils import logging as reqlog
+from .utils.metrics import Metrics
+
+
+class RawScoreServer:
+    def __init__(self, cfg: ServeConfig):
+        self.cfg = cfg
+        self.metrics = Metrics()
+        self.engines = []
+        self.batchers = []
+        self.drift_sync = None
+        self._rr = 0
+
+    # ------------------------------------------------------------ lifecycle
+    async def start(self):
+        from .parallel import DriftSync
+        from .serve import _build_engines
+
+        cfg = self.cfg
+        self.engines = _build_engines(cfg)
+        self.drift_sync = DriftSync(self.engines[0].packed, device="cpu")
+
+        def scorer(e):
+            def run(codes, nums):
+                out = e.score_arrays(codes, nums)
+                self._fold_drift(out, nums)
+                return out
+
+            return run
+
+        def scorer_single(e):
+            if e.device != "cuda":
+                return None
+
+            def run(codes, nums):
+                out = e.score_encoded_bytes(codes, nums)
+                self._fold_drift(out, nums)
+                return out
+
+            return run
+
+        self.batchers = [
+            MicroBatcher(
+                scorer(e),
+                max_rows=cfg.max_batch_rows,
+                max_wait_us=cfg.batch_wait_us,
+                score_single=scorer_single(e),
+            )
+            for e in self.engines
+        ]
+        for b in self.batchers:
+            await b.start()
+        self._server = await asyncio.start_server(
+            self._handle, cfg.host, cfg.port, backlog=512
+        )
+        return self._server
+
+    async def close(self):
+        self._server.close()
+        await self._server.wait_closed()
+        for b in self.batchers:
+            await b.close()
+
+    def _fold_drift(self, out, nums):
+        if "cat_hist" in out:
+            self.drift_sync.accumulate(out["cat_hist"], nums)
+            if self.drift_sync.batches % max(self.cfg.drift_sync_period, 1) == 0:
+                self.drift_sync.allreduce()
+                self.metrics.observe_drift_sync()
+
+    # ------------------------------------------------------------ handlers
+    async def _score(self, body: bytes) -> tuple[int, bytes]:
+        cfg = self.cfg
+        request_id = uuid.uuid4().hex
+        if cfg.log_inference_data:
+            reqlog.log_inference_data(
+                cfg.service_name, request_id, body.decode("utf-8", "replace")
+            )
+        engine = self.engines[self._rr % len(self.engines)]
+        self._rr += 1
+        loop = asyncio.get_running_loop()
+        try:
+            codes, nums = await loop.run_in_executor(
+                None, engine.encode_json_body, body
+            )
+        except (ValueError, TypeError):
+            # full pydantic validation for reference 422/coercion semantics
+            from pydantic import TypeAdapter, ValidationError
+
+            from .pack import encode_batch
+
+            try:
+                data = TypeAdapter(list[LoanApplicant]).validate_json(body)
+            except ValidationError as e:
+                return 422, json.dumps({"detail": e.errors(include_url=False)}).encode()
+            codes, nums = encode_batch([r.__dict__ for r in data], engine.packed.vocabs)
+        if len(codes) == 0:
+            return 400, b'{"detail": "empty request batch"}'
+
+        idx = (self._rr - 1) % len(self.batchers)
+        t0 = time.perf_counter()
+        try:
+            out = await self.batchers[idx].submit(codes, nums)
+        except Exception as e:
+            self.metrics.observe_error()
+            return 500, json.dumps({"detail": f"scoring failed: {e}"}).encode()
+        latency_ms = (time.perf_counter() - t0) * 1e3
+        self.metrics.observe_request(len(codes), latency_ms)
+
+        if "response_bytes" in out:
+            rb = out["response_bytes"]
+            reqlog.log_model_output_raw(
+                cfg.service_name, request_id, rb.decode("utf-8", "replace"),
+                latency_ms=latency_ms, rows=out["rows"],
+                device=f"{engine.device}:{engine.device_index}",
+            )
+            return 200, rb
+        one_minus = (
+            np.float32(1.0) - np.asarray(out["p_vals"], dtype=np.float32)
+        ).astype(np.float64)
+        response = {
+            "predictions": np.asarray(out["predictions"]).tolist(),
+            "outliers": np.asarray(out["outliers"]).tolist(),
+            "feature_drift_batch": dict(zip(FEATURES, one_minus.tolist())),
+        }
+        payload = json.dumps(response).encode()
+        reqlog.log_model_output(
+            cfg.service_name, request_id, response,
+            latency_ms=latency_ms, rows=len(codes),
+            device=f"{engine.device}:{engine.device_index}",
+        )
+        return 200, payload
+
+    async def _get(self, path: bytes) -> tuple[int, bytes]:
+        if path == b"/healthz":
+            return 200, json.dumps(
+                {"status": "ok", "engines": len(self.engines)}
+            ).encode()
+        if path == b"/metrics":
+            return 200, json.dumps(self.metrics.snapshot()).encode()
+        if path == b"/drift":
+            self.drift_sync.allreduce()
+            return 200, json.dumps(self.drift_sync.snapshot()).encode()
+        return 404, b'{"detail": "not found"}'
+
+    # ------------------------------------------------------------ HTTP/1.1
+    async def _handle(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
+        try:
+            while True:
+                line = await reader.readline()
+                if not line or line == b"\r\n":
+                    break
+                try:
+                    method, path, _ = line.split(b" ", 2)
+                except ValueError:
+                    break
+                clen = 0
+                keep_alive = True
+                while True:
+                    h = await reader.readline()
+                    if h in (b"\r\n", b"\n", b""):
+                        break
+                    k, _, v = h.partition(b":")
+                    lk = k.lower()
+                    if lk == b"content-length":
+                        clen = int(v.strip())
+                    elif lk == b"connection" and b"close" in v.lower():
+                        keep_alive = False
+                    elif lk == b"transfer-encoding":
+                        # chunked bodies: not needed by the contract clients
+                        await self._respond(writer, 411, b'{"detail": "length required"}')
+                        return
+                body = await reader.readexactly(clen) if clen else b""
+                if method == b"POST" and path in (b"/score", b"/predict"):
+                    status, payload = await self._score(body)
+                elif method == b"GET":
+                    status, payload = await self._get(path)
+                else:
+                    status, payload = 405, b'{"detail": "method not allowed"}'
+                await self._respond(writer, status, payload)
+                if not keep_alive:
+                    break
+        except (asyncio.IncompleteReadError, ConnectionResetError):
+            pass
+        finally:
+            try:
+                writer.close()
+                await writer.wait_closed()
+            except Exception:
+                pass
+
+    _REASONS = {200: b"OK", 400: b"Bad Request", 404: b"Not Found",
+                405: b"Method Not Allowed", 411: b"Length Required",
+                422: b"Unprocessable Entity", 500: b"Internal Server Error"}
+
+    async def _respond(self, writer, status: int, payload: bytes):
+        head = b"HTTP/1.1 %d %s\r\nContent-Type: application/json\r\nContent-Length: %d\r\n\r\n" % (
+            status, self._REASONS.get(status, b"OK"), len(payload),
+        )
+        writer.write(head + payload)
+        await writer.drain()
+
+
+async def serve_raw(cfg: ServeConfig):
+    server = RawScoreServer(cfg)
+    await server.start()
+    print(f"[rawserve] listening on {cfg.host}:{cfg.port}", flush=True)
+    async with server._server:
+        await server._server.serve_forever()
+
+
+def main(cfg: ServeConfig):
+    asyncio.run(serve_raw(cfg))

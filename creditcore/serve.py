@@ -350,6 +350,12 @@ def main(argv: list[str] | None = None):
     logging.basicConfig(level=logging.INFO)
     cfg = ServeConfig.from_args(argv)
     app = create_app(cfg)
+    if cfg.raw_http:
+        # throughput frontend: minimal asyncio HTTP/1.1, native JSON path
+        from .rawserve import main as raw_main
+
+        raw_main(cfg)
+        return
     # access_log off: the app writes its own two JSON lines per request
     if cfg.workers > 1:
         # process-level scaling: each worker owns its engines and GIL

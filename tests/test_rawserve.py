@@ -1,0 +1,96 @@
+"""Raw asyncio HTTP frontend: contract parity with the FastAPI app on the
+hot endpoints (creditcore.rawserve)."""
+
+from __future__ import annotations
+
+import asyncio
+import json
+
+import pytest
+
+
+@pytest.fixture(scope="module")
+def raw_url(model_dir):
+    import socket
+    import threading
+
+    from creditcore.config import ServeConfig
+    from creditcore.rawserve import RawScoreServer
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    cfg.host = "127.0.0.1"
+    cfg.port = port
+
+    loop = asyncio.new_event_loop()
+    server = RawScoreServer(cfg)
+    started = threading.Event()
+
+    def run():
+        asyncio.set_event_loop(loop)
+
+        async def go():
+            await server.start()
+            started.set()
+            await server._server.serve_forever()
+
+        try:
+            loop.run_until_complete(go())
+        except asyncio.CancelledError:
+            pass
+
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    assert started.wait(timeout=120)
+    yield f"http://127.0.0.1:{port}"
+    loop.call_soon_threadsafe(loop.stop)
+
+
+def test_score_roundtrip(raw_url):
+    import httpx
+
+    from creditcore.schema import SAMPLE_REQUEST, ModelOutput
+
+    r = httpx.post(f"{raw_url}/score", json=SAMPLE_REQUEST, timeout=60.0)
+    assert r.status_code == 200
+    ModelOutput.model_validate(r.json())
+    r2 = httpx.post(f"{raw_url}/predict", json=SAMPLE_REQUEST, timeout=60.0)
+    assert r2.json()["predictions"] == r.json()["predictions"]
+
+
+def test_batch_and_keepalive(raw_url):
+    import httpx
+
+    from creditcore.data import make_request_batch
+
+    with httpx.Client(timeout=60.0) as c:  # keep-alive across requests
+        for seed in (1, 2):
+            r = c.post(f"{raw_url}/score", json=make_request_batch(64, seed=seed))
+            assert r.status_code == 200
+            assert len(r.json()["predictions"]) == 64
+
+
+def test_error_paths(raw_url):
+    import httpx
+
+    assert httpx.post(f"{raw_url}/score", content=b"[]").status_code == 400
+    assert httpx.post(f"{raw_url}/score", json=[{"sex": None}]).status_code == 422
+    assert httpx.get(f"{raw_url}/nope").status_code == 404
+    assert httpx.get(f"{raw_url}/healthz").json()["status"] == "ok"
+    r = httpx.post(f"{raw_url}/score", json=[{"credit_limit": "18000"}])
+    assert r.status_code == 200  # pydantic lax fallback
+
+
+def test_metrics_and_drift(raw_url):
+    import httpx
+
+    m = httpx.get(f"{raw_url}/metrics").json()
+    assert m["requests_total"] >= 1
+    d = httpx.get(f"{raw_url}/drift").json()
+    assert len(d["node_feature_drift"]) == 23
