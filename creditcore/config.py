@@ -59,7 +59,13 @@ class ServeConfig:
             flag = "--" + f.name.replace("_", "-")
             cur = getattr(cfg, f.name)
             if isinstance(cur, bool):
-                p.add_argument(flag, type=lambda s: s.lower() in ("1", "true", "yes"), default=cur)
+                p.add_argument(
+                    flag,
+                    type=lambda s: s.lower() in ("1", "true", "yes"),
+                    nargs="?",
+                    const=True,
+                    default=cur,
+                )
             else:
                 p.add_argument(flag, type=type(cur), default=cur)
         ns = p.parse_args(argv)
