@@ -50,6 +50,7 @@ class ServeConfig:
 
     # observability
     log_inference_data: bool = field(default_factory=lambda: _env("log_inference_data", True, bool))
+    log_responses: bool = field(default_factory=lambda: _env("log_responses", True, bool))
 
     @classmethod
     def from_args(cls, argv: list[str] | None = None) -> "ServeConfig":

@@ -212,7 +212,8 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         if "response_bytes" in out:  # solo-flush wire-out fast path
             rb = out["response_bytes"]
             metrics.observe_request(out["rows"], latency_ms)
-            reqlog.log_model_output_raw(
+            if cfg.log_responses:
+                reqlog.log_model_output_raw(
                 cfg.service_name,
                 request_id,
                 rb.decode("utf-8", "replace"),
