@@ -137,11 +137,12 @@ class RawScoreServer:
 
         if "response_bytes" in out:
             rb = out["response_bytes"]
-            reqlog.log_model_output_raw(
-                cfg.service_name, request_id, rb.decode("utf-8", "replace"),
-                latency_ms=latency_ms, rows=out["rows"],
-                device=f"{engine.device}:{engine.device_index}",
-            )
+            if cfg.log_responses:
+                reqlog.log_model_output_raw(
+                    cfg.service_name, request_id, rb.decode("utf-8", "replace"),
+                    latency_ms=latency_ms, rows=out["rows"],
+                    device=f"{engine.device}:{engine.device_index}",
+                )
             return 200, rb
         one_minus = (
             np.float32(1.0) - np.asarray(out["p_vals"], dtype=np.float32)
@@ -152,11 +153,12 @@ class RawScoreServer:
             "feature_drift_batch": dict(zip(FEATURES, one_minus.tolist())),
         }
         payload = json.dumps(response).encode()
-        reqlog.log_model_output(
-            cfg.service_name, request_id, response,
-            latency_ms=latency_ms, rows=len(codes),
-            device=f"{engine.device}:{engine.device_index}",
-        )
+        if cfg.log_responses:
+            reqlog.log_model_output(
+                cfg.service_name, request_id, response,
+                latency_ms=latency_ms, rows=len(codes),
+                device=f"{engine.device}:{engine.device_index}",
+            )
         return 200, payload
 
     async def _get(self, path: bytes) -> tuple[int, bytes]:

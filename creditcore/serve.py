@@ -214,13 +214,13 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
             metrics.observe_request(out["rows"], latency_ms)
             if cfg.log_responses:
                 reqlog.log_model_output_raw(
-                cfg.service_name,
-                request_id,
-                rb.decode("utf-8", "replace"),
-                latency_ms=latency_ms,
-                rows=out["rows"],
-                device=f"{engines[idx].device}:{engines[idx].device_index}",
-            )
+                    cfg.service_name,
+                    request_id,
+                    rb.decode("utf-8", "replace"),
+                    latency_ms=latency_ms,
+                    rows=out["rows"],
+                    device=f"{engines[idx].device}:{engines[idx].device_index}",
+                )
             return rb
 
         one_minus = (
@@ -232,14 +232,15 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
             "feature_drift_batch": dict(zip(FEATURES, one_minus.tolist())),
         }
         metrics.observe_request(len(codes), latency_ms)
-        reqlog.log_model_output(
-            cfg.service_name,
-            request_id,
-            response,
-            latency_ms=latency_ms,
-            rows=len(codes),
-            device=f"{engines[idx].device}:{engines[idx].device_index}",
-        )
+        if cfg.log_responses:
+            reqlog.log_model_output(
+                cfg.service_name,
+                request_id,
+                response,
+                latency_ms=latency_ms,
+                rows=len(codes),
+                device=f"{engines[idx].device}:{engines[idx].device_index}",
+            )
         return response
 
     # The endpoints take the raw body (native JSON fast path) but keep the
