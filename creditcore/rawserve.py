@@ -79,7 +79,9 @@ class RawScoreServer:
         for b in self.batchers:
             await b.start()
         self._server = await asyncio.start_server(
-            self._handle, cfg.host, cfg.port, backlog=512
+            self._handle, cfg.host, cfg.port, backlog=512,
+            reuse_port=(cfg.workers > 1)
+            or __import__("os").environ.get("CREDITCORE_RAW_REUSE_PORT") == "1",
         )
         return self._server
 
@@ -240,4 +242,53 @@ async def serve_raw(cfg: ServeConfig):
 
 
 def main(cfg: ServeConfig):
+    if cfg.workers > 1:
+        _spawn_workers(cfg)
+        return
     asyncio.run(serve_raw(cfg))
+
+
+def _spawn_workers(cfg: ServeConfig):
+    """Production multi-GPU shape: N worker processes share the port via
+    SO_REUSEPORT (the kernel is the request queue), each pinned to one GPU
+    through HIP_VISIBLE_DEVICES — the reference's K8s-replica scaling
+    collapsed into one node."""
+    import dataclasses
+    import os
+    import signal
+    import subprocess
+    import sys
+
+    env_base = dict(os.environ)
+    for f in dataclasses.fields(cfg):
+        env_base[f"CREDITCORE_{f.name.upper()}"] = str(getattr(cfg, f.name))
+    env_base["CREDITCORE_WORKERS"] = "1"  # workers serve single-process
+    try:
+        import torch
+
+        n_dev = torch.cuda.device_count()
+    except Exception:
+        n_dev = 0
+
+    procs = []
+    for w in range(cfg.workers):
+        env = dict(env_base)
+        if n_dev > 0:
+            env["HIP_VISIBLE_DEVICES"] = str(w % n_dev)
+            env["CREDITCORE_N_GPUS"] = "1"
+        env["CREDITCORE_RAW_HTTP"] = "1"
+        env["CREDITCORE_RAW_REUSE_PORT"] = "1"
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "creditcore", "serve"], env=env
+        ))
+    print(f"[rawserve] {cfg.workers} workers on port {cfg.port} "
+          f"(SO_REUSEPORT, one GPU each)", flush=True)
+
+    def shutdown(*_):
+        for pr in procs:
+            pr.terminate()
+
+    signal.signal(signal.SIGTERM, shutdown)
+    signal.signal(signal.SIGINT, shutdown)
+    for pr in procs:
+        pr.wait()
