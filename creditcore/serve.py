@@ -276,16 +276,40 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         return _json_response(await _predict_impl(await request.body()))
 
     @app.get("/healthz")
-    async def healthz():
+    async def healthz(deep: bool = False):
         pool: ReplicaPool = state["pool"]
         any_alive = any(pool.alive)
-        return {
+        body = {
             "status": "ok" if any_alive else "dead",
             "engines": [
                 {"device": e.device, "index": e.device_index, "alive": pool.alive[i]}
                 for i, e in enumerate(state["engines"])
             ],
         }
+        if deep and any_alive:
+            # active probe: score the schema-default record on every live
+            # replica (SURVEY.md §5.3 — HIP error surfacing on demand)
+            import asyncio
+
+            from .schema import LoanApplicant
+
+            codes, nums = encode_batch(
+                [LoanApplicant().__dict__], state["engines"][0].packed.vocabs
+            )
+            loop = asyncio.get_running_loop()
+            for i, e in enumerate(state["engines"]):
+                if not pool.alive[i]:
+                    continue
+                try:
+                    await loop.run_in_executor(
+                        None, lambda e=e: e.score_arrays(codes, nums, False)
+                    )
+                    body["engines"][i]["probe"] = "ok"
+                except Exception as exc:
+                    pool.report_fail(i)
+                    body["engines"][i]["probe"] = f"failed: {exc}"
+                    body["status"] = "degraded"
+        return body
 
     @app.get("/metrics")
     async def metrics_endpoint():

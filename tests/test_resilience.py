@@ -84,3 +84,9 @@ def test_engine_failure_marks_replica(client, monkeypatch):
         assert r.status_code == 500
     assert state["pool"].alive == [False]
     assert client.post("/predict", json=SAMPLE_REQUEST).status_code == 503
+
+
+def test_deep_health_probe(client):
+    body = client.get("/healthz?deep=1").json()
+    assert body["status"] == "ok"
+    assert body["engines"][0]["probe"] == "ok"
