@@ -164,6 +164,7 @@ class RawScoreServer:
         return 200, payload
 
     async def _get(self, path: bytes) -> tuple[int, bytes]:
+        path = path.partition(b"?")[0]
         if path == b"/healthz":
             return 200, json.dumps(
                 {"status": "ok", "engines": len(self.engines)}
@@ -203,7 +204,7 @@ class RawScoreServer:
                         await self._respond(writer, 411, b'{"detail": "length required"}')
                         return
                 body = await reader.readexactly(clen) if clen else b""
-                if method == b"POST" and path in (b"/score", b"/predict"):
+                if method == b"POST" and path.partition(b"?")[0] in (b"/score", b"/predict"):
                     status, payload = await self._score(body)
                 elif method == b"GET":
                     status, payload = await self._get(path)
