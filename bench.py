@@ -39,14 +39,16 @@ def _build_packed(cache_dir: str, seed: int = 2024):
     )
     from creditcore.schema import FEATURES, TARGET
 
+    algo = BENCH_MODEL.get("algo", "rf")
     path = os.path.join(
         cache_dir,
-        f"bench_packed_{BENCH_MODEL['n_estimators']}x{BENCH_MODEL['max_depth']}_{TRAIN_ROWS}.npz",
+        f"bench_packed_{algo}_{BENCH_MODEL['n_estimators']}x{BENCH_MODEL['max_depth']}_{TRAIN_ROWS}.npz",
     )
     if os.path.exists(path):
         return PackedModel.load(path), path
     df = make_uci_shaped_frame(n_rows=TRAIN_ROWS, seed=seed)
-    pipe = make_classifier_pipeline({**BENCH_MODEL, "random_state": seed})
+    params = {k: v for k, v in BENCH_MODEL.items() if k != "algo"}
+    pipe = make_classifier_pipeline({**params, "random_state": seed}, algo)
     pipe.fit(df[FEATURES], df[TARGET].values.ravel())
     drift, outlier = T.fit_detectors(df)
     c = pack_classifier_pipeline(pipe)
@@ -70,11 +72,14 @@ def main():
     p.add_argument("--model-trees", type=int, default=None,
                    help="override bench model n_estimators (sensitivity runs)")
     p.add_argument("--model-depth", type=int, default=None)
+    p.add_argument("--model-algo", default="rf", choices=["rf", "gbt"],
+                   help="classifier family for the bench model")
     args = p.parse_args()
     if args.model_trees:
         BENCH_MODEL["n_estimators"] = args.model_trees
     if args.model_depth:
         BENCH_MODEL["max_depth"] = args.model_depth
+    BENCH_MODEL["algo"] = args.model_algo
 
     import torch
 
@@ -273,7 +278,8 @@ def main():
                     "data": "synthetic (UCI-credit-default-shaped), random-seed-fitted model",
                     "config": {
                         "model": (
-                            f"RandomForest {BENCH_MODEL['n_estimators']}x"
+                            f"{'GBT' if BENCH_MODEL.get('algo') == 'gbt' else 'RandomForest'} "
+                            f"{BENCH_MODEL['n_estimators']}x"
                             f"depth{BENCH_MODEL['max_depth']} + IForest100 + TabularDrift"
                         ),
                         "global_batch": n_gpus * args.rows,

@@ -1,20 +1,20 @@
 """Scoring engine — one model replica on one device.
 
-The GPU path uploads the packed flat buffers (creditcore.pack) to HBM once and
-scores request batches with the HIP kernels (csrc/creditcore_kernels.hip) on a private
-HIP stream with pinned staging buffers:
+The GPU path holds a C++ ScoreSession (csrc/creditcore_kernels.hip):
+resident HBM model buffers, two private HIP streams, double-buffered pinned
+staging, and one hipGraph per recurring request shape. A request is:
 
-    host:   encode strings -> codes (int16), nums (f32)      [request parse]
-    H2D:    codes + nums via pinned staging                  [engine stream]
-    GPU:    score_pipeline kernel set:
-              forest traversal  -> P(default) per row
-              iforest traversal -> instance score + outlier flag
-              drift             -> per-cat histograms + per-num K-S D
-    D2H:    proba/outlier/stat buffers (small)
-    host:   chi2 / K-S statistic -> p-value (scipy, 23 scalars)
+    host:   native JSON parse -> codes (int16), nums (f32)   [C, GIL-free]
+    graph:  pinned H2D -> accumulator memset -> 2-tree-ILP forest kernels
+            (classifier + isolation forest) ∥ drift branch on stream 2
+            (categorical histogram + 512-thread K-S) -> finalize -> D2H
+    host:   drift p-values (C: chi2 closed forms + MTW/Pelz-Good K-S sf)
+            -> response JSON bytes (C, shortest-round-trip doubles)
 
-Everything between H2D and D2H is asynchronous on the engine stream; the only
-sync is the final D2H, which is part of request latency anyway.
+Entry points by use: score_json_full (single C++ call, lowest latency),
+submit_encoded_slot/finish_slot (double-buffered pipelining: step i's
+epilogue overlaps step i+1's graph), score_arrays (array in/out for the
+micro-batcher), score_records/score_json (dict/response-shaped outputs).
 
 The CPU path (device="cpu") uses the same packed buffers via
 creditcore.ops.cpu_ref — identical numerics (the HIP kernels are tested

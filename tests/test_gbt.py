@@ -72,3 +72,30 @@ def test_gbt_gpu_parity(gbt_packed, score_batch):
     nums_imp = cpu_ref.impute_nums(packed, nums)
     ref = cpu_ref.score_forest_cpu(packed, codes, nums_imp)
     np.testing.assert_allclose(out["predictions"], ref, atol=1e-9)
+
+
+def test_gbt_serves_through_api(train_df, tmp_path):
+    """A GBT-trained pyfunc dir serves through the same API stack."""
+    from fastapi.testclient import TestClient
+
+    from creditcore import registry
+    from creditcore.config import ServeConfig
+    from creditcore.schema import SAMPLE_REQUEST
+    from creditcore.serve import create_app
+    from creditcore.train import fit_detectors
+
+    pipe = make_classifier_pipeline(
+        {"n_estimators": 40, "max_depth": 3, "random_state": 1}, algorithm="gbt"
+    )
+    pipe.fit(train_df[FEATURES].head(1500), train_df[TARGET].head(1500).values.ravel())
+    drift, outlier = fit_detectors(train_df.head(1500))
+    model_dir = str(tmp_path / "gbt_model")
+    registry.save_pyfunc_model(model_dir, pipe, drift, outlier)
+
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    with TestClient(create_app(cfg)) as client:
+        r = client.post("/predict", json=SAMPLE_REQUEST)
+        assert r.status_code == 200
+        assert 0.0 <= r.json()["predictions"][0] <= 1.0
