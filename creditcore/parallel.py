@@ -233,3 +233,44 @@ class DriftSync:
         """In-process multi-GPU merge (no process group): sum another
         replica's local histogram into this one (cross-GPU copy over xGMI)."""
         self.local += other.local.to(self.local.device)
+
+    # -------------------------------------------------- cross-process merge
+    # SO_REUSEPORT serving workers are separate processes with no process
+    # group; they publish local histograms to a shared directory (tmpfs)
+    # and /drift sums every worker's latest snapshot.
+    def publish(self, publish_dir: str) -> None:
+        import os
+        import tempfile
+
+        os.makedirs(publish_dir, exist_ok=True)
+        arr = self.local.cpu().numpy()
+        fd, tmp = tempfile.mkstemp(dir=publish_dir, suffix=".tmp")
+        try:
+            with os.fdopen(fd, "wb") as f:
+                np.save(f, arr)
+            os.replace(tmp, os.path.join(publish_dir, f"drift_{os.getpid()}.npy"))
+        except BaseException:
+            try:
+                os.unlink(tmp)
+            except OSError:
+                pass
+            raise
+
+    def merge_published(self, publish_dir: str) -> None:
+        """Set global_ to the sum of every published worker snapshot (this
+        worker's live local histogram replaces its own stale file)."""
+        import glob
+        import os
+
+        import torch
+
+        total = self.local.cpu().clone()
+        me = f"drift_{os.getpid()}.npy"
+        for f in glob.glob(os.path.join(publish_dir, "drift_*.npy")):
+            if os.path.basename(f) == me:
+                continue
+            try:
+                total += torch.from_numpy(np.load(f))
+            except (OSError, ValueError):
+                continue  # mid-replace or removed
+        self.global_ = total.to(self.local.device)

@@ -91,11 +91,20 @@ class RawScoreServer:
         for b in self.batchers:
             await b.close()
 
+    @property
+    def _drift_dir(self) -> str:
+        import os
+        import tempfile
+
+        base = "/dev/shm" if os.path.isdir("/dev/shm") else tempfile.gettempdir()
+        return os.path.join(base, f"creditcore_drift_{self.cfg.port}")
+
     def _fold_drift(self, out, nums):
         if "cat_hist" in out:
             self.drift_sync.accumulate(out["cat_hist"], nums)
             if self.drift_sync.batches % max(self.cfg.drift_sync_period, 1) == 0:
                 self.drift_sync.allreduce()
+                self.drift_sync.publish(self._drift_dir)
                 self.metrics.observe_drift_sync()
 
     # ------------------------------------------------------------ handlers
@@ -172,7 +181,9 @@ class RawScoreServer:
         if path == b"/metrics":
             return 200, json.dumps(self.metrics.snapshot()).encode()
         if path == b"/drift":
-            self.drift_sync.allreduce()
+            # node-global: this worker's live state + every worker's
+            # published snapshot (SO_REUSEPORT workers share no memory)
+            self.drift_sync.merge_published(self._drift_dir)
             return 200, json.dumps(self.drift_sync.snapshot()).encode()
         return 404, b'{"detail": "not found"}'
 

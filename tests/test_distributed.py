@@ -160,3 +160,35 @@ def test_driftsync_detects_node_drift(packed):
     drift = ds.snapshot()["node_feature_drift"]
     assert drift["credit_limit"] > 0.99  # shifted numerics
     assert drift["sex"] > 0.99  # all-female batch vs mixed reference
+
+
+def test_driftsync_cross_process_publish(packed, tmp_path):
+    """SO_REUSEPORT workers share drift via published snapshots: two
+    instances publishing to the same dir see each other's counts."""
+    import os
+
+    import torch
+
+    from creditcore.data import make_request_batch
+    from creditcore.ops import cpu_ref
+    from creditcore.pack import encode_batch
+    from creditcore.parallel import DriftSync
+
+    d = str(tmp_path / "shm")
+    a = DriftSync(packed, device="cpu", n_bins=8)
+    b = DriftSync(packed, device="cpu", n_bins=8)
+    for ds, seed in ((a, 1), (b, 2)):
+        recs = make_request_batch(60, seed=seed)
+        codes, nums = encode_batch(recs, packed.vocabs)
+        nums_imp = cpu_ref.impute_nums(packed, nums)
+        hist, _ = cpu_ref.drift_stats_cpu(packed, codes, nums_imp)
+        ds.accumulate(torch.from_numpy(hist), torch.from_numpy(nums))
+    # emulate two pids: publish b under a different name
+    b.publish(d)
+    os.replace(
+        os.path.join(d, f"drift_{os.getpid()}.npy"),
+        os.path.join(d, "drift_99999.npy"),
+    )
+    a.merge_published(d)
+    snap = a.snapshot()
+    assert snap["rows"] == 120  # both workers' numeric rows visible
