@@ -206,3 +206,28 @@ def test_serving_fallback_handles_lax_types(model_dir):
         assert r.status_code == 200
         r = client.post("/predict", json=[{"sex": None}])
         assert r.status_code == 422
+
+
+def test_c_serializer_double_roundtrip(ext):
+    """append_double (C, shortest-round-trip) must reproduce every double
+    bit-exactly through JSON, including awkward magnitudes."""
+    import json as _json
+
+    import torch
+
+    vals = [0.0, 1.0, 0.1, 2 / 3, 1e-300, 1.7976931348623157e308,
+            5e-324, 123456789.123456789, 0.49999999999999994, 1e16, 1e-5]
+    b = len(vals)
+    outs = torch.zeros(3 * b, dtype=torch.float64)
+    outs[:b] = torch.tensor(vals, dtype=torch.float64)
+    outs[2 * b:] = torch.tensor([i % 2 for i in range(b)], dtype=torch.float64)
+    pvals = np.linspace(0.0, 1.0, 23)
+    from creditcore.schema import FEATURES
+
+    raw = ext.build_response_json(outs, b, np.ascontiguousarray(pvals), FEATURES)
+    doc = _json.loads(raw)
+    for got, want in zip(doc["predictions"], vals):
+        assert got == want, (got, want)
+    assert doc["outliers"] == [float(i % 2) for i in range(b)]
+    for f, p in zip(FEATURES, pvals):
+        assert doc["feature_drift_batch"][f] == float(np.float32(1.0) - np.float32(p))
