@@ -168,8 +168,8 @@ def main():
         from collections import deque
         from concurrent.futures import ThreadPoolExecutor
 
-        executor = ThreadPoolExecutor(max_workers=2)
-        DEPTH = 3  # encode i+1..i+3 overlap scoring of i (C parser drops the GIL)
+        executor = ThreadPoolExecutor(max_workers=3)
+        DEPTH = 4  # encode i+1..i+4 overlap scoring of i (C parser drops the GIL)
 
         use_slots = device == "cuda" and args.rows <= 16384
 
