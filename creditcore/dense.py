@@ -276,13 +276,10 @@ class DenseEngine:
         if with_drift:
             db = min(b, self.MAX_DRIFT_ROWS)
             if db > 16384:
-                # large batches: impute+transpose to [F, B] (contiguous
-                # per-feature rows), rocPRIM segmented sort along rows, then
-                # the chip-filling scan kernel (F x row-chunk blocks,
-                # atomicMax) with fully coalesced accesses
-                xi_t = torch.where(
-                    torch.isnan(xt[:db].t()), g["medians"][:, None], xt[:db].t()
-                )
+                # large batches: fused impute+transpose kernel -> [F, B],
+                # rocPRIM segmented sort along rows, then the chip-filling
+                # scan kernel (F x row-chunk blocks, atomicMax)
+                xi_t = g["ext"].impute_transpose(xt[:db].contiguous(), g["medians"])
                 xs = xi_t.sort(dim=1).values
                 ks_d = g["ext"].ks_stats_sorted(xs.contiguous(), g["ref_sorted"], g["rs_off"])
             else:

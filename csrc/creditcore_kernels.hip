@@ -666,6 +666,57 @@ std::vector<torch::Tensor> dense_score(
   return {proba, iscore, outlier};
 }
 
+// Fused impute + transpose for the large-batch dense drift path:
+// x [B, F] row-major -> xt [F, B] row-major with NaN -> median, one pass
+// through an LDS tile (32x32, +1 padding column against bank conflicts).
+__global__ __launch_bounds__(256) void impute_transpose_kernel(
+    const float* __restrict__ x,   // [B, F]
+    const float* __restrict__ medians,  // [F]
+    int n_rows,
+    int n_feat,
+    float* __restrict__ xt)        // [F, B]
+{
+  __shared__ float tile[32][33];
+  const int f0 = blockIdx.x * 32;
+  const int r0 = blockIdx.y * 32;
+  const int tx = threadIdx.x & 31;   // feature within tile on load
+  const int ty = threadIdx.x >> 5;   // row group (8 rows per pass)
+#pragma unroll
+  for (int k = 0; k < 4; ++k) {
+    const int r = r0 + ty + k * 8;
+    const int f = f0 + tx;
+    float v = 0.0f;
+    if (r < n_rows && f < n_feat) {
+      v = x[(size_t)r * n_feat + f];
+      if (isnan(v)) v = medians[f];
+    }
+    tile[ty + k * 8][tx] = v;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int k = 0; k < 4; ++k) {
+    const int f = f0 + ty + k * 8;   // feature on store
+    const int r = r0 + tx;           // row on store (coalesced)
+    if (f < n_feat && r < n_rows) xt[(size_t)f * n_rows + r] = tile[tx][ty + k * 8];
+  }
+}
+
+torch::Tensor impute_transpose(torch::Tensor x, torch::Tensor medians) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kFloat32 && x.is_contiguous());
+  const int B = (int)x.size(0);
+  const int F = (int)x.size(1);
+  TORCH_CHECK((int)medians.numel() == F, "median size mismatch");
+  auto xt = torch::empty({F, B},
+      torch::TensorOptions().dtype(torch::kFloat32).device(x.device()));
+  hipStream_t stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(impute_transpose_kernel,
+      dim3(ceil_div(F, 32), ceil_div(B, 32)), dim3(256), 0, stream,
+      x.data_ptr<float>(), medians.data_ptr<float>(), B, F,
+      xt.data_ptr<float>());
+  HIP_CHECK(hipGetLastError());
+  return xt;
+}
+
 // Scan-only K-S for pre-sorted batch columns (large-B dense path: the sort
 // goes through rocPRIM (torch.sort), then this kernel fills the chip with
 // (feature x row-chunk) blocks; per-feature max via f32-bit atomicMax).
@@ -1856,6 +1907,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       });
   m.def("dense_score", &dense_score,
         "Fused impute + logistic linear score + robust-z outlier (gfx950)");
+  m.def("impute_transpose", &impute_transpose,
+        "Fused NaN->median impute + transpose [B,F] -> [F,B] (gfx950)");
   m.def("ks_stats_sorted", &ks_stats_sorted,
         "Exact K-S D per column for pre-sorted batch columns (gfx950)");
   m.def("forest_ilp_bench", &forest_ilp_bench,
