@@ -20,12 +20,16 @@ Layout:
     train.py     — local trainer (reference 01-train-model.ipynb)
     registry.py  — MLflow pyfunc artifact layout (reference 02-register cell-12)
     pack.py      — pickles -> flat GPU buffers (tree SoA, vocabs, ECDFs)
-    models/      — CPU reference detectors (forest, iforest, drift, linear)
-    ops/         — HIP kernel wrappers (csrc/ extension)
-    engine.py    — GPU scoring engine (HIP streams, pinned staging)
+    models/      — CPU reference detectors (forest/GBT, iforest, drift, linear)
+    ops/         — HIP kernel wrappers (csrc/ extension) + CPU golden reference
+    engine.py    — GPU scoring engine over the C++ ScoreSession (hipGraphs,
+                   pinned slots, native JSON in/out)
     batching.py  — micro-batch request aggregator
-    parallel/    — RCCL replica group: weight broadcast + drift all-reduce
-    serve.py     — FastAPI app: POST /predict (+ /score alias)
+    parallel.py  — RCCL weight broadcast + node-global drift sync
+    serve.py     — FastAPI app: /predict /score /healthz /metrics /drift
+    rawserve.py  — raw asyncio HTTP frontend (SO_REUSEPORT workers)
+    dense.py     — 10M x 1k wide-tabular family (HBM-resident drift refs)
+    pipeline.py  — local CD pipeline (train -> staging -> smoke -> prod)
 """
 
 __version__ = "0.1.0"
