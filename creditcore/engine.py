@@ -309,14 +309,18 @@ class ScoringEngine:
 
 
 def load_engine(model_directory: str, device: str = "auto", device_index: int = 0) -> ScoringEngine:
-    """Load a pyfunc model dir (or packed .npz) into a ScoringEngine."""
+    """Load a model into a ScoringEngine. Accepts a pyfunc model dir, a
+    packed .npz, or a registry URI (``models:/<name>/<version|latest>`` —
+    the reference's MLflow registry addressing, 02-register cell-15)."""
     import os
 
     from . import pack as packmod
     from .config import ServeConfig
+    from .registry import resolve_model_uri
 
     if device == "auto":
         device = ServeConfig().resolve_device()
+    model_directory = resolve_model_uri(model_directory)
     if model_directory.endswith(".npz") and os.path.isfile(model_directory):
         packed = PackedModel.load(model_directory)
     else:

@@ -122,3 +122,17 @@ def test_resolve_plain_path_passthrough():
 def test_load_pyfunc_rejects_non_model_dir(tmp_path):
     with pytest.raises(FileNotFoundError):
         registry.load_pyfunc_model(str(tmp_path))
+
+
+def test_serve_from_registry_uri(model_dir, tmp_path, monkeypatch):
+    """load_engine resolves models:/name/latest (registry-addressed serving,
+    the reference's models:/<name>/<version> contract)."""
+    from creditcore.engine import load_engine
+    from creditcore.schema import SAMPLE_REQUEST
+
+    root = str(tmp_path / "reg")
+    registry.register_model(model_dir, "credit-default-uci-custom", root)
+    monkeypatch.setattr(registry, "DEFAULT_REGISTRY_ROOT", root)
+    eng = load_engine("models:/credit-default-uci-custom/latest", device="cpu")
+    out = eng.score_records(SAMPLE_REQUEST)
+    assert 0.0 <= out["response"]["predictions"][0] <= 1.0
