@@ -216,13 +216,14 @@ DEFAULT_REGISTRY_ROOT = os.environ.get("CREDITCORE_REGISTRY", "./registry")
 def register_model(
     model_dir: str,
     name: str,
-    registry_root: str = DEFAULT_REGISTRY_ROOT,
+    registry_root: str | None = None,
     tags: dict | None = None,
 ) -> str:
     """Copy/record a pyfunc model dir into the local registry; returns the
     model URI ``models:/<name>/<version>`` (reference cell-13/15)."""
     import shutil
 
+    registry_root = registry_root or DEFAULT_REGISTRY_ROOT
     base = os.path.join(registry_root, name)
     os.makedirs(base, exist_ok=True)
     versions = [int(v) for v in os.listdir(base) if v.isdigit()]
@@ -241,11 +242,12 @@ def register_model(
     return f"models:/{name}/{version}"
 
 
-def resolve_model_uri(uri: str, registry_root: str = DEFAULT_REGISTRY_ROOT) -> str:
+def resolve_model_uri(uri: str, registry_root: str | None = None) -> str:
     """Resolve ``models:/<name>/<version|latest>`` or a plain path to a model
     directory."""
     if not uri.startswith("models:/"):
         return uri
+    registry_root = registry_root or DEFAULT_REGISTRY_ROOT
     rest = uri[len("models:/") :]
     name, _, version = rest.partition("/")
     base = os.path.join(registry_root, name)
