@@ -338,8 +338,12 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         if rows == 0 or len(body) != expect:
             raise HTTPException(status_code=422, detail="body size mismatch")
         x = np.frombuffer(body, dtype="<f4", offset=8).reshape(rows, cols)
+        import asyncio
+
         t0 = time.perf_counter()
-        out = eng.score_arrays(x)
+        out = await asyncio.get_running_loop().run_in_executor(
+            None, eng.score_arrays, x
+        )
         latency_ms = (time.perf_counter() - t0) * 1e3
         state["metrics"].observe_request(rows, latency_ms)
         return _json_response(
