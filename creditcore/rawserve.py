@@ -25,6 +25,7 @@ import numpy as np
 
 from .batching import MicroBatcher
 from .config import ServeConfig
+from .serve import ReplicaPool
 from .schema import FEATURES, LoanApplicant
 from .utils import logging as reqlog
 from .utils.metrics import Metrics
@@ -37,7 +38,7 @@ class RawScoreServer:
         self.engines = []
         self.batchers = []
         self.drift_sync = None
-        self._rr = 0
+        self.pool = None
 
     # ------------------------------------------------------------ lifecycle
     async def start(self):
@@ -78,6 +79,7 @@ class RawScoreServer:
         ]
         for b in self.batchers:
             await b.start()
+        self.pool = ReplicaPool(len(self.engines))
         self._server = await asyncio.start_server(
             self._handle, cfg.host, cfg.port, backlog=512,
             reuse_port=(cfg.workers > 1)
@@ -115,8 +117,12 @@ class RawScoreServer:
             reqlog.log_inference_data(
                 cfg.service_name, request_id, body.decode("utf-8", "replace")
             )
-        engine = self.engines[self._rr % len(self.engines)]
-        self._rr += 1
+        try:
+            idx = self.pool.pick()
+        except RuntimeError:
+            self.metrics.observe_error()
+            return 503, b'{"detail": "no healthy replicas"}'
+        engine = self.engines[idx]
         loop = asyncio.get_running_loop()
         try:
             codes, nums = await loop.run_in_executor(
@@ -136,12 +142,13 @@ class RawScoreServer:
         if len(codes) == 0:
             return 400, b'{"detail": "empty request batch"}'
 
-        idx = (self._rr - 1) % len(self.batchers)
         t0 = time.perf_counter()
         try:
             out = await self.batchers[idx].submit(codes, nums)
+            self.pool.report_ok(idx)
         except Exception as e:
             self.metrics.observe_error()
+            self.pool.report_fail(idx)
             return 500, json.dumps({"detail": f"scoring failed: {e}"}).encode()
         latency_ms = (time.perf_counter() - t0) * 1e3
         self.metrics.observe_request(len(codes), latency_ms)
@@ -175,8 +182,13 @@ class RawScoreServer:
     async def _get(self, path: bytes) -> tuple[int, bytes]:
         path = path.partition(b"?")[0]
         if path == b"/healthz":
+            any_alive = any(self.pool.alive)
             return 200, json.dumps(
-                {"status": "ok", "engines": len(self.engines)}
+                {
+                    "status": "ok" if any_alive else "dead",
+                    "engines": len(self.engines),
+                    "alive": self.pool.alive,
+                }
             ).encode()
         if path == b"/metrics":
             return 200, json.dumps(self.metrics.snapshot()).encode()
@@ -235,7 +247,8 @@ class RawScoreServer:
 
     _REASONS = {200: b"OK", 400: b"Bad Request", 404: b"Not Found",
                 405: b"Method Not Allowed", 411: b"Length Required",
-                422: b"Unprocessable Entity", 500: b"Internal Server Error"}
+                422: b"Unprocessable Entity", 500: b"Internal Server Error",
+                503: b"Service Unavailable"}
 
     async def _respond(self, writer, status: int, payload: bytes):
         head = b"HTTP/1.1 %d %s\r\nContent-Type: application/json\r\nContent-Length: %d\r\n\r\n" % (
