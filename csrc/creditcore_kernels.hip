@@ -845,7 +845,7 @@ struct ScoreSession {
   int device_index{};
   hipStream_t stream{};
   hipStream_t stream2{};  // drift branch (runs parallel to the forests)
-  hipEvent_t ev_fork{}, ev_join{}, ev_mid{};
+  hipEvent_t ev_fork{}, ev_join{};
   hipEvent_t ev_done[2]{};  // per-slot completion (async score)
 
   // raw slot pointers into the pinned buffers
@@ -915,7 +915,6 @@ struct ScoreSession {
     HIP_CHECK(hipEventCreateWithFlags(&ev_join, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_done[0], hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_done[1], hipEventDisableTiming));
-    HIP_CHECK(hipEventCreateWithFlags(&ev_mid, hipEventDisableTiming));
     HIP_CHECK(hipDeviceSynchronize());  // uploads above used torch's stream
   }
 
@@ -929,7 +928,6 @@ struct ScoreSession {
       (void)hipEventDestroy(ev_join);
       (void)hipEventDestroy(ev_done[0]);
       (void)hipEventDestroy(ev_done[1]);
-      (void)hipEventDestroy(ev_mid);
     }
   }
 
@@ -941,12 +939,14 @@ struct ScoreSession {
         (size_t)b * N_CAT * sizeof(short), hipMemcpyHostToDevice, stream));
     HIP_CHECK(hipMemcpyAsync(d_nums.data_ptr(), p_nums(slot),
         (size_t)b * N_NUM * sizeof(float), hipMemcpyHostToDevice, stream));
-    // fork point: stream2 work depends only on the H2D copies + memset
+    // fork point: the drift branch (stream2) depends only on the H2D copies
+    if (with_drift) {
+      HIP_CHECK(hipEventRecord(ev_fork, stream));
+      HIP_CHECK(hipStreamWaitEvent(stream2, ev_fork, 0));
+    }
     double* acc_cls = acc.data_ptr<double>();
     double* acc_if = acc_cls + b;  // b-packed: one memset clears both
     HIP_CHECK(hipMemsetAsync(acc_cls, 0, (size_t)(2 * b) * sizeof(double), stream));
-    HIP_CHECK(hipEventRecord(ev_fork, stream));
-    HIP_CHECK(hipStreamWaitEvent(stream2, ev_fork, 0));
 
     const int row_blocks = ceil_div(b, BLOCK);
     // 2-tree-ILP traversal measured faster at every batch size
@@ -955,34 +955,18 @@ struct ScoreSession {
     auto chunks = [&](int64_t t) {
       return std::max(1, std::min(ceil_div(2048, row_blocks), (int)((t + 1) / 2)));
     };
-    // The classifier forest is split across the two streams (~2/3 + ~1/3)
-    // so its latency-bound traversal overlaps itself as well as the drift
-    // branch; acc updates are atomic, finalize waits on ev_mid.
-    const int64_t t_split = with_drift ? (2 * t_cls) / 3 : t_cls;
-    if (t_split > 0)
-      hipLaunchKernelGGL((forest_kernel_ilp<false>), dim3(row_blocks, chunks(t_split)),
-          dim3(BLOCK), 0, stream,
-          d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
-          reinterpret_cast<const int4*>(cls_nodes.data_ptr<int>()),
-          cls_off.data_ptr<int>(), (int)t_split,
-          feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), b, acc_cls);
-    if (t_split < t_cls)
-      hipLaunchKernelGGL((forest_kernel_ilp<false>), dim3(row_blocks, chunks(t_cls - t_split)),
-          dim3(BLOCK), 0, stream2,
-          d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
-          reinterpret_cast<const int4*>(cls_nodes.data_ptr<int>()),
-          cls_off.data_ptr<int>() + t_split, (int)(t_cls - t_split),
-          feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), b, acc_cls);
+    hipLaunchKernelGGL((forest_kernel_ilp<false>), dim3(row_blocks, chunks(t_cls)),
+        dim3(BLOCK), 0, stream,
+        d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
+        reinterpret_cast<const int4*>(cls_nodes.data_ptr<int>()),
+        cls_off.data_ptr<int>(), (int)t_cls,
+        feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), b, acc_cls);
     hipLaunchKernelGGL((forest_kernel_ilp<true>), dim3(row_blocks, chunks(t_if)),
         dim3(BLOCK), 0, stream,
         d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
         reinterpret_cast<const int4*>(if_nodes.data_ptr<int>()),
         if_off.data_ptr<int>(), (int)t_if, nullptr, nullptr, b, acc_if);
 
-    if (t_split < t_cls) {  // finalize needs stream2's forest share
-      HIP_CHECK(hipEventRecord(ev_mid, stream2));
-      HIP_CHECK(hipStreamWaitEvent(stream, ev_mid, 0));
-    }
     double* proba = outs.data_ptr<double>();
     hipLaunchKernelGGL(finalize_kernel, dim3(row_blocks), dim3(BLOCK), 0, stream,
         acc_cls, acc_if, b, cls_kind, 1.0 / (double)t_cls, cls_bias,
