@@ -13,6 +13,7 @@ import pytest
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an AMD GPU (run on MI355X)")
+    config.addinivalue_line("markers", "slow: multi-process / lifecycle tests")
 
 
 def _has_gpu() -> bool:

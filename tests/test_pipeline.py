@@ -19,6 +19,7 @@ def _free_port():
     return port
 
 
+@pytest.mark.slow
 @pytest.mark.timeout(300)
 def test_full_pipeline(tmp_path):
     report = run_pipeline(
