@@ -303,8 +303,17 @@ def _spawn_workers(cfg: ServeConfig):
             env["CREDITCORE_N_GPUS"] = "1"
         env["CREDITCORE_RAW_HTTP"] = "1"
         env["CREDITCORE_RAW_REUSE_PORT"] = "1"
+        def _die_with_parent():  # orphan guard (Linux PR_SET_PDEATHSIG)
+            import ctypes
+
+            try:
+                ctypes.CDLL("libc.so.6").prctl(1, signal.SIGTERM)
+            except OSError:  # pragma: no cover
+                pass
+
         procs.append(subprocess.Popen(
-            [sys.executable, "-m", "creditcore", "serve"], env=env
+            [sys.executable, "-m", "creditcore", "serve"], env=env,
+            preexec_fn=_die_with_parent,
         ))
     print(f"[rawserve] {cfg.workers} workers on port {cfg.port} "
           f"(SO_REUSEPORT, one GPU each)", flush=True)
