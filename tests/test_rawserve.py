@@ -42,7 +42,8 @@ def raw_url(model_dir):
 
         try:
             loop.run_until_complete(go())
-        except asyncio.CancelledError:
+        except (asyncio.CancelledError, RuntimeError):
+            # loop.stop() from the teardown aborts run_until_complete
             pass
 
     t = threading.Thread(target=run, daemon=True)
@@ -50,6 +51,7 @@ def raw_url(model_dir):
     assert started.wait(timeout=120)
     yield f"http://127.0.0.1:{port}"
     loop.call_soon_threadsafe(loop.stop)
+    t.join(timeout=30)
 
 
 def test_score_roundtrip(raw_url):
