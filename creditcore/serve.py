@@ -379,9 +379,10 @@ def main(argv: list[str] | None = None):
 
     logging.basicConfig(level=logging.INFO)
     cfg = ServeConfig.from_args(argv)
-    app = create_app(cfg)
     if cfg.raw_http:
-        # throughput frontend: minimal asyncio HTTP/1.1, native JSON path
+        # throughput frontend: minimal asyncio HTTP/1.1, native JSON path.
+        # No create_app here: a multi-worker parent must not load engines
+        # (each worker owns its own GPU + engine).
         from .rawserve import main as raw_main
 
         raw_main(cfg)
@@ -404,7 +405,10 @@ def main(argv: list[str] | None = None):
             workers=cfg.workers,
         )
     else:
-        uvicorn.run(app, host=cfg.host, port=cfg.port, log_level="info", access_log=False)
+        uvicorn.run(
+            create_app(cfg), host=cfg.host, port=cfg.port,
+            log_level="info", access_log=False,
+        )
 
 
 if __name__ == "__main__":

@@ -98,6 +98,97 @@ def test_metrics_and_drift(raw_url):
     assert len(d["node_feature_drift"]) == 23
 
 
+def test_chunked_body_gets_411(raw_url):
+    """Transfer-Encoding: chunked is outside the contract clients' subset —
+    the frontend must refuse it deterministically, not misparse it."""
+    import socket
+
+    host, port = raw_url[len("http://"):].rsplit(":", 1)
+    with socket.create_connection((host, int(port)), timeout=15) as s:
+        s.sendall(
+            b"POST /score HTTP/1.1\r\nHost: x\r\n"
+            b"Transfer-Encoding: chunked\r\n\r\n"
+        )
+        data = s.recv(4096)
+    assert data.startswith(b"HTTP/1.1 411")
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(180)
+def test_multi_worker_reuseport_lifecycle(model_dir, tmp_path):
+    """Production shape on CPU: 2 SO_REUSEPORT workers behind one port.
+    Both must serve through the shared socket, and the PDEATHSIG orphan
+    guard must take the workers down when the parent dies uncleanly."""
+    import os
+    import signal
+    import socket
+    import subprocess
+    import sys
+    import time
+
+    import httpx
+    import psutil
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+
+    import creditcore
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(creditcore.__file__)))
+    env = dict(os.environ, CREDITCORE_LOG_INFERENCE_DATA="0",
+               CREDITCORE_LOG_RESPONSES="0",
+               PYTHONPATH=repo + os.pathsep + os.environ.get("PYTHONPATH", ""))
+    parent = subprocess.Popen(
+        [sys.executable, "-m", "creditcore", "serve", "--raw-http",
+         "--workers", "2", "--device", "cpu", "--host", "127.0.0.1",
+         "--port", str(port), "--model-directory", model_dir],
+        env=env, cwd=str(tmp_path),
+        stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+    )
+    workers = []
+    try:
+        from creditcore.schema import SAMPLE_REQUEST
+
+        deadline = time.monotonic() + 120
+        ok = 0
+        while time.monotonic() < deadline and ok < 4:
+            workers = workers or psutil.Process(parent.pid).children(recursive=True)
+            try:
+                r = httpx.post(f"http://127.0.0.1:{port}/score",
+                               json=SAMPLE_REQUEST, timeout=10.0)
+                if r.status_code == 200:
+                    ok += 1
+                    continue
+            except Exception:
+                pass
+            time.sleep(1.0)
+        assert ok >= 4, "workers never served through the shared port"
+
+        workers = psutil.Process(parent.pid).children(recursive=True)
+        assert len(workers) >= 2
+        # unclean parent death → PDEATHSIG must reap the workers
+        os.kill(parent.pid, signal.SIGKILL)
+        parent.wait(timeout=30)
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline and any(w.is_running() for w in workers):
+            time.sleep(0.5)
+        assert not any(w.is_running() for w in workers), "orphaned workers survived"
+    finally:
+        # exact handles only (never pattern-kill): the spawned parent and
+        # the worker PIDs captured while it was alive
+        try:
+            parent.kill()
+        except Exception:
+            pass
+        for w in workers:
+            try:
+                w.kill()
+            except Exception:
+                pass
+
+
 def test_raw_replica_failure_gives_503(raw_url, model_dir):
     """Raw frontend drops a repeatedly-failing replica like the FastAPI app."""
     import httpx
