@@ -168,8 +168,12 @@ def main():
         from collections import deque
         from concurrent.futures import ThreadPoolExecutor
 
-        executor = ThreadPoolExecutor(max_workers=3)
-        DEPTH = 4  # encode i+1..i+4 overlap scoring of i (C parser drops the GIL)
+        # A/B-tunable pipeline shape (defaults chosen by measurement on
+        # MI355X; see profiles/kernel_tuning.md)
+        enc_workers = int(os.environ.get("CREDITCORE_BENCH_ENCODE_WORKERS", "3"))
+        executor = ThreadPoolExecutor(max_workers=enc_workers)
+        # encode i+1..i+DEPTH overlap scoring of i (C parser drops the GIL)
+        DEPTH = int(os.environ.get("CREDITCORE_BENCH_DEPTH", "4"))
 
         use_slots = device == "cuda" and args.rows <= 16384 and with_drift
 
