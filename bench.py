@@ -170,10 +170,10 @@ def main():
 
         # A/B-tunable pipeline shape (defaults chosen by measurement on
         # MI355X; see profiles/kernel_tuning.md)
-        enc_workers = int(os.environ.get("CREDITCORE_BENCH_ENCODE_WORKERS", "3"))
+        enc_workers = int(os.environ.get("CREDITCORE_BENCH_ENCODE_WORKERS", "4"))
         executor = ThreadPoolExecutor(max_workers=enc_workers)
         # encode i+1..i+DEPTH overlap scoring of i (C parser drops the GIL)
-        DEPTH = int(os.environ.get("CREDITCORE_BENCH_DEPTH", "4"))
+        DEPTH = int(os.environ.get("CREDITCORE_BENCH_DEPTH", "6"))
 
         use_slots = device == "cuda" and args.rows <= 16384 and with_drift
 
