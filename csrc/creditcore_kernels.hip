@@ -836,8 +836,10 @@ torch::Tensor ks_stats(
 struct ScoreSession {
   torch::Tensor cls_nodes, cls_off, feat_col, feat_code, medians;
   torch::Tensor if_nodes, if_off, ref_sorted, rs_off, cat_off;
-  torch::Tensor d_codes, d_nums, acc, outs, hist, ksd;
-  torch::Tensor pin_codes, pin_nums, pin_outs, pin_hist, pin_ksd;
+  torch::Tensor d_codes, d_nums, acc, outs, d_drift;
+  size_t drift_bytes = 0, drift_ksd_off = 0;  // [hist i32 | ksd f32] blob
+  torch::Tensor pin_codes, pin_nums, pin_outs, pin_drift;
+  torch::Tensor pin_hist, pin_ksd;  // typed views into pin_drift
   double if_denom{}, if_offset{}, if_threshold{};
   int cls_kind{};
   double cls_bias{};
@@ -852,8 +854,11 @@ struct ScoreSession {
   int16_t* p_codes(int s) { return pin_codes.data_ptr<int16_t>() + (size_t)s * capacity * N_CAT; }
   float* p_nums(int s) { return pin_nums.data_ptr<float>() + (size_t)s * capacity * N_NUM; }
   double* p_outs(int s) { return pin_outs.data_ptr<double>() + (size_t)s * 3 * capacity; }
-  int32_t* p_hist(int s) { return pin_hist.data_ptr<int32_t>() + (size_t)s * total_bins; }
-  float* p_ksd(int s) { return pin_ksd.data_ptr<float>() + (size_t)s * N_NUM; }
+  uint8_t* p_drift(int s) { return pin_drift.data_ptr<uint8_t>() + (size_t)s * drift_bytes; }
+  int32_t* p_hist(int s) { return reinterpret_cast<int32_t*>(p_drift(s)); }
+  float* p_ksd(int s) { return reinterpret_cast<float*>(p_drift(s) + drift_ksd_off); }
+  int32_t* d_hist() { return reinterpret_cast<int32_t*>(d_drift.data_ptr<uint8_t>()); }
+  float* d_ksd() { return reinterpret_cast<float*>(d_drift.data_ptr<uint8_t>() + drift_ksd_off); }
 
   ScoreSession(py::dict model, int64_t cap, int dev) : capacity(cap), device_index(dev) {
     c10::hip::HIPGuard guard((c10::DeviceIndex)dev);
@@ -888,8 +893,11 @@ struct ScoreSession {
     d_nums = torch::empty({capacity, N_NUM}, devopt.dtype(torch::kFloat32));
     acc = torch::empty({2, capacity}, devopt.dtype(torch::kFloat64));
     outs = torch::empty({3, capacity}, devopt.dtype(torch::kFloat64));
-    hist = torch::empty({total_bins}, devopt.dtype(torch::kInt32));
-    ksd = torch::empty({N_NUM}, devopt.dtype(torch::kFloat32));
+    // drift outputs packed into one blob: [hist i32 | ksd f32] — both are
+    // b-independent sizes, so one D2H covers the whole drift branch
+    drift_ksd_off = (size_t)total_bins * sizeof(int32_t);
+    drift_bytes = drift_ksd_off + (size_t)N_NUM * sizeof(float);
+    d_drift = torch::empty({(int64_t)drift_bytes}, devopt.dtype(torch::kUInt8));
 
     // two slots: step i's host epilogue reads slot i%2 while step i+1's
     // graph fills the other slot (pipelined serving/bench loops)
@@ -897,8 +905,10 @@ struct ScoreSession {
     pin_codes = torch::empty({2, capacity, N_CAT}, pinned.dtype(torch::kInt16));
     pin_nums = torch::empty({2, capacity, N_NUM}, pinned.dtype(torch::kFloat32));
     pin_outs = torch::empty({2, 3, capacity}, pinned.dtype(torch::kFloat64));
-    pin_hist = torch::empty({2, total_bins}, pinned.dtype(torch::kInt32));
-    pin_ksd = torch::empty({2, N_NUM}, pinned.dtype(torch::kFloat32));
+    pin_drift = torch::empty({2, (int64_t)drift_bytes}, pinned.dtype(torch::kUInt8));
+    pin_hist = pin_drift.slice(1, 0, (int64_t)drift_ksd_off).view(torch::kInt32);
+    pin_ksd = pin_drift.slice(1, (int64_t)drift_ksd_off, (int64_t)drift_bytes)
+                  .view(torch::kFloat32);
 
     auto rs = py::cast<torch::Tensor>(model["ref_sorted_offsets"]);
     auto rs_acc = rs.accessor<int32_t, 1>();
@@ -977,12 +987,12 @@ struct ScoreSession {
       // Drift branch forked onto stream2 right after the H2D copies: the
       // K-S sort+scan and categorical histogram overlap the forest kernels
       // (captured as parallel graph branches; joined before the end).
-      HIP_CHECK(hipMemsetAsync(hist.data_ptr(), 0, (size_t)total_bins * sizeof(int), stream2));
+      HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), stream2));
       const int hist_blocks = std::min(row_blocks, 1024);
       hipLaunchKernelGGL(cat_hist_kernel, dim3(hist_blocks), dim3(BLOCK),
           (size_t)total_bins * sizeof(int), stream2,
           d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
-          hist.data_ptr<int>());
+          d_hist());
       int m_pow2 = 512 / 64;  // >= one cross-wave reduction slot per wave
       while (m_pow2 < b) m_pow2 <<= 1;
       // Stage the reference column in LDS when batch + ref fit the 160 KiB
@@ -1000,16 +1010,18 @@ struct ScoreSession {
           batch_bytes + (ref_lds ? ref_bytes : 0), stream2,
           d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b, m_pow2,
           (int)ref_lds, ref_sorted.data_ptr<float>(),
-          rs_off.data_ptr<int64_t>(), ksd.data_ptr<float>());
-      HIP_CHECK(hipMemcpyAsync(p_hist(slot), hist.data_ptr(),
-          (size_t)total_bins * sizeof(int), hipMemcpyDeviceToHost, stream2));
-      HIP_CHECK(hipMemcpyAsync(p_ksd(slot), ksd.data_ptr(),
-          (size_t)N_NUM * sizeof(float), hipMemcpyDeviceToHost, stream2));
+          rs_off.data_ptr<int64_t>(), d_ksd());
+      // one D2H for the whole drift branch (hist + K-S D share a blob)
+      HIP_CHECK(hipMemcpyAsync(p_drift(slot), d_drift.data_ptr<uint8_t>(),
+          drift_bytes, hipMemcpyDeviceToHost, stream2));
       HIP_CHECK(hipEventRecord(ev_join, stream2));
-      HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
     }
+    // classifier-output D2H depends only on finalize — it overlaps the
+    // drift branch's K-S tail; the join lands after it so graph completion
+    // still covers both streams
     HIP_CHECK(hipMemcpyAsync(p_outs(slot), proba,
         (size_t)(3 * b) * sizeof(double), hipMemcpyDeviceToHost, stream));
+    if (with_drift) HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
     HIP_CHECK(hipGetLastError());
   }
 
