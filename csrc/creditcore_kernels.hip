@@ -437,6 +437,88 @@ __global__ __launch_bounds__(BS) void ks_kernel_t(
   }
 }
 
+// Small-batch K-S variant: instead of a bitonic sort (log^2(B)/2 barriered
+// LDS passes — 55 stages at B=1024, ~39 us with only n_cols workgroups on
+// the chip), count each element's strict / non-strict rank directly with an
+// O(B^2) LDS sweep. The inner loop reads s_vals[k] at the same k across the
+// whole wavefront (LDS broadcast, no bank conflicts, no barriers), so at
+// B<=2048 the quadratic sweep is several times faster than the sort's
+// barrier chain. Output is bitwise-identical to ks_kernel_t: lt/le ARE the
+// sorted path's tie-run lower/upper bounds, and the ref binary searches and
+// f64 CDF arithmetic are the same (models/drift.ks_2samp_d semantics).
+template <int BS>
+__global__ __launch_bounds__(BS) void ks_count_kernel_t(
+    const float* __restrict__ nums,       // [B, n_cols]
+    const float* __restrict__ medians,    // [n_cols]
+    int n_cols, int n_rows,
+    const float* __restrict__ ref_sorted, const int64_t* __restrict__ rs_off,
+    float* __restrict__ ks_d)
+{
+  extern __shared__ float s_vals[];  // [n_rows] imputed batch column
+  const int j = blockIdx.x;
+  const int m = n_rows;
+  for (int i = threadIdx.x; i < m; i += blockDim.x) {
+    float v = nums[i * n_cols + j];
+    if (isnan(v)) v = medians[j];
+    s_vals[i] = v;
+  }
+  const int64_t ref_lo = rs_off[j];
+  const int n = (int)(rs_off[j + 1] - ref_lo);
+  const float* __restrict__ ref = ref_sorted + ref_lo;
+  __syncthreads();
+
+  double dmax = 0.0;
+  for (int i = threadIdx.x; i < m; i += blockDim.x) {
+    const float x = s_vals[i];
+    int lt = 0, le = 0;  // # batch elements < x / <= x
+    int k = 0;
+    for (; k + 3 < m; k += 4) {
+      const float v0 = s_vals[k], v1 = s_vals[k + 1];
+      const float v2 = s_vals[k + 2], v3 = s_vals[k + 3];
+      lt += (int)(v0 < x) + (int)(v1 < x) + (int)(v2 < x) + (int)(v3 < x);
+      le += (int)(v0 <= x) + (int)(v1 <= x) + (int)(v2 <= x) + (int)(v3 <= x);
+    }
+    for (; k < m; ++k) {
+      const float v = s_vals[k];
+      lt += (int)(v < x);
+      le += (int)(v <= x);
+    }
+    int l = 0, r = n;  // lower_bound in ref
+    while (l < r) {
+      const int mid = (l + r) >> 1;
+      if (ref[mid] < x) l = mid + 1; else r = mid;
+    }
+    const int sl = l;
+    r = n;  // upper_bound in ref (resume from sl)
+    while (l < r) {
+      const int mid = (l + r) >> 1;
+      if (ref[mid] <= x) l = mid + 1; else r = mid;
+    }
+    const int sr = l;
+    const double fl = fabs((double)sl / n - (double)lt / m);
+    const double fr = fabs((double)sr / n - (double)le / m);
+    dmax = fmax(dmax, fmax(fl, fr));
+  }
+
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    dmax = fmax(dmax, __shfl_down(dmax, off, 64));
+  __syncthreads();
+  float* s_red = s_vals;  // one slot per wave (launch smem floors at BS/64)
+  const int wave = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) s_red[wave] = (float)dmax;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float d = s_red[0];
+    for (int w = 1; w < (int)(blockDim.x >> 6); ++w) d = fmaxf(d, s_red[w]);
+    ks_d[j] = d;
+  }
+}
+
+// counting beats the sort up to here (O(B^2/threads) vs the bitonic's
+// barrier chain; crossover measured ~2k on MI355X)
+constexpr int KS_COUNT_MAX_ROWS = 2048;
+
 // ---------------------------------------------------------------------------
 // Host launchers
 // ---------------------------------------------------------------------------
@@ -577,15 +659,24 @@ std::vector<torch::Tensor> drift_stats(
       codes.data_ptr<short>(), B, cat_off.data_ptr<int>(), (int)total_bins,
       hist.data_ptr<int>());
 
-  int m_pow2 = 1;
-  while (m_pow2 < B) m_pow2 <<= 1;
-  // LDS must also hold one cross-wave reduction slot per wave
-  m_pow2 = std::max(m_pow2, BLOCK / 64);
-  hipLaunchKernelGGL((ks_kernel_t<256>), dim3(N_NUM), dim3(BLOCK),
-      (size_t)m_pow2 * sizeof(float), stream,
-      nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, B, m_pow2,
-      /*ref_lds=*/0, ref_sorted.data_ptr<float>(),
-      rs_off64.data_ptr<int64_t>(), ks_d.data_ptr<float>());
+  if (B <= KS_COUNT_MAX_ROWS) {
+    const size_t smem = (size_t)std::max(B, BLOCK / 64) * sizeof(float);
+    hipLaunchKernelGGL((ks_count_kernel_t<256>), dim3(N_NUM), dim3(BLOCK),
+        smem, stream,
+        nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, B,
+        ref_sorted.data_ptr<float>(), rs_off64.data_ptr<int64_t>(),
+        ks_d.data_ptr<float>());
+  } else {
+    int m_pow2 = 1;
+    while (m_pow2 < B) m_pow2 <<= 1;
+    // LDS must also hold one cross-wave reduction slot per wave
+    m_pow2 = std::max(m_pow2, BLOCK / 64);
+    hipLaunchKernelGGL((ks_kernel_t<256>), dim3(N_NUM), dim3(BLOCK),
+        (size_t)m_pow2 * sizeof(float), stream,
+        nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, B, m_pow2,
+        /*ref_lds=*/0, ref_sorted.data_ptr<float>(),
+        rs_off64.data_ptr<int64_t>(), ks_d.data_ptr<float>());
+  }
   HIP_CHECK(hipGetLastError());
 
   return {hist, ks_d};
@@ -993,24 +1084,28 @@ struct ScoreSession {
           (size_t)total_bins * sizeof(int), stream2,
           d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
           d_hist());
-      int m_pow2 = 512 / 64;  // >= one cross-wave reduction slot per wave
-      while (m_pow2 < b) m_pow2 <<= 1;
-      // Stage the reference column in LDS when batch + ref fit the 160 KiB
-      // CU budget; searches then stay on-chip.
-      const size_t batch_bytes = (size_t)m_pow2 * sizeof(float);
-      const size_t ref_bytes = (size_t)max_ref_len * sizeof(float);
-      // measured (bench/kernel_micro.py on MI355X): staging the ref column
-      // in LDS is slower at b=1024 (62 vs 48 us) and within noise at 16k;
-      // ks_kernel is __launch_bounds__(256), so block stays 256.
-      (void)ref_bytes;
-      const bool ref_lds = false;
-      // 512 threads: halves the bitonic's serial depth per thread
-      // (kernel_micro: 48.6->31.2 us @1k, 963->509 @16k)
-      hipLaunchKernelGGL((ks_kernel_t<512>), dim3(N_NUM), dim3(512),
-          batch_bytes + (ref_lds ? ref_bytes : 0), stream2,
-          d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b, m_pow2,
-          (int)ref_lds, ref_sorted.data_ptr<float>(),
-          rs_off.data_ptr<int64_t>(), d_ksd());
+      if (b <= KS_COUNT_MAX_ROWS) {
+        // O(B^2) counting path: no sort, no barrier chain (38.6 -> single-
+        // digit us at b=1024; see profiles/kernel_tuning.md)
+        const size_t smem = (size_t)std::max(b, 512 / 64) * sizeof(float);
+        hipLaunchKernelGGL((ks_count_kernel_t<512>), dim3(N_NUM), dim3(512),
+            smem, stream2,
+            d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b,
+            ref_sorted.data_ptr<float>(), rs_off.data_ptr<int64_t>(), d_ksd());
+      } else {
+        int m_pow2 = 512 / 64;  // >= one cross-wave reduction slot per wave
+        while (m_pow2 < b) m_pow2 <<= 1;
+        // measured (bench/kernel_micro.py on MI355X): staging the ref
+        // column in LDS is slower at b=1024 (62 vs 48 us) and within noise
+        // at 16k, so the ref stays in L2 (ref_lds=0).
+        // 512 threads: halves the bitonic's serial depth per thread
+        // (kernel_micro: 48.6->31.2 us @1k, 963->509 @16k)
+        hipLaunchKernelGGL((ks_kernel_t<512>), dim3(N_NUM), dim3(512),
+            (size_t)m_pow2 * sizeof(float), stream2,
+            d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b,
+            m_pow2, /*ref_lds=*/0, ref_sorted.data_ptr<float>(),
+            rs_off.data_ptr<int64_t>(), d_ksd());
+      }
       // one D2H for the whole drift branch (hist + K-S D share a blob)
       HIP_CHECK(hipMemcpyAsync(p_drift(slot), d_drift.data_ptr<uint8_t>(),
           drift_bytes, hipMemcpyDeviceToHost, stream2));
