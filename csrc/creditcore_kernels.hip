@@ -515,9 +515,11 @@ __global__ __launch_bounds__(BS) void ks_count_kernel_t(
   }
 }
 
-// counting beats the sort up to here (O(B^2/threads) vs the bitonic's
-// barrier chain; crossover measured ~2k on MI355X)
-constexpr int KS_COUNT_MAX_ROWS = 2048;
+// counting beats the sort only while the quadratic sweep stays tiny:
+// measured on MI355X, counting is 51 us at B=1024 vs the bitonic's 38.6 us
+// (the LDS-broadcast sweep is ALU-bound, not barrier-bound), but wins
+// clearly on the small batches the single-call latency path serves
+constexpr int KS_COUNT_MAX_ROWS = 256;
 
 // ---------------------------------------------------------------------------
 // Host launchers
