@@ -263,8 +263,8 @@ __global__ __launch_bounds__(BLOCK) void forest_kernel_ilpN(
 }
 
 __global__ __launch_bounds__(BLOCK) void finalize_kernel(
-    const double* __restrict__ cls_acc,
-    const double* __restrict__ if_acc,
+    double* __restrict__ cls_acc,
+    double* __restrict__ if_acc,
     int n_rows,
     int cls_kind,        // 0 = RF leaf-fraction mean; 1 = GBT logit sum
     double inv_n_trees,
@@ -286,6 +286,12 @@ __global__ __launch_bounds__(BLOCK) void finalize_kernel(
   const double s = anomaly + if_offset;
   iscore[i] = s;
   outlier[i] = (s > if_threshold) ? 1.0 : 0.0;
+  // zero-after-read: the accumulators start all-zero (torch::zeros at
+  // session init) and every finalize re-zeros the rows it consumed, so no
+  // per-request memset node is needed in the graph regardless of how batch
+  // sizes interleave across captured shapes
+  cls_acc[i] = 0.0;
+  if_acc[i] = 0.0;
 }
 
 // ---------------------------------------------------------------------------
@@ -320,8 +326,15 @@ __global__ __launch_bounds__(BLOCK) void cat_hist_kernel(
     }
   }
   __syncthreads();
-  for (int i = threadIdx.x; i < total_bins; i += blockDim.x)
-    if (s_hist[i] != 0) atomicAdd(&hist[i], s_hist[i]);
+  if (gridDim.x == 1) {
+    // single-block launch (small batches): plain stores — the output needs
+    // no pre-zero memset node in the graph
+    for (int i = threadIdx.x; i < total_bins; i += blockDim.x)
+      hist[i] = s_hist[i];
+  } else {
+    for (int i = threadIdx.x; i < total_bins; i += blockDim.x)
+      if (s_hist[i] != 0) atomicAdd(&hist[i], s_hist[i]);
+  }
 }
 
 // One block per numeric feature. Sorts the (imputed) batch column in LDS with
@@ -984,7 +997,9 @@ struct ScoreSession {
 
     d_codes = torch::empty({capacity, N_CAT}, devopt.dtype(torch::kInt16));
     d_nums = torch::empty({capacity, N_NUM}, devopt.dtype(torch::kFloat32));
-    acc = torch::empty({2, capacity}, devopt.dtype(torch::kFloat64));
+    // all-zero invariant: finalize_kernel re-zeros the rows it consumes,
+    // so record() never needs a memset node for the accumulators
+    acc = torch::zeros({2, capacity}, devopt.dtype(torch::kFloat64));
     outs = torch::empty({3, capacity}, devopt.dtype(torch::kFloat64));
     // drift outputs packed into one blob: [hist i32 | ksd f32] — both are
     // b-independent sizes, so one D2H covers the whole drift branch
@@ -1048,8 +1063,7 @@ struct ScoreSession {
       HIP_CHECK(hipStreamWaitEvent(stream2, ev_fork, 0));
     }
     double* acc_cls = acc.data_ptr<double>();
-    double* acc_if = acc_cls + b;  // b-packed: one memset clears both
-    HIP_CHECK(hipMemsetAsync(acc_cls, 0, (size_t)(2 * b) * sizeof(double), stream));
+    double* acc_if = acc_cls + b;  // b-packed; all-zero by invariant
 
     const int row_blocks = ceil_div(b, BLOCK);
     // 2-tree-ILP traversal measured faster at every batch size
@@ -1080,8 +1094,11 @@ struct ScoreSession {
       // Drift branch forked onto stream2 right after the H2D copies: the
       // K-S sort+scan and categorical histogram overlap the forest kernels
       // (captured as parallel graph branches; joined before the end).
-      HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), stream2));
-      const int hist_blocks = std::min(row_blocks, 1024);
+      // small batches: one block overwrites the histogram (no memset node);
+      // larger ones pre-zero + atomically accumulate across blocks
+      const int hist_blocks = (b <= 2048) ? 1 : std::min(row_blocks, 1024);
+      if (hist_blocks > 1)
+        HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), stream2));
       hipLaunchKernelGGL(cat_hist_kernel, dim3(hist_blocks), dim3(BLOCK),
           (size_t)total_bins * sizeof(int), stream2,
           d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
