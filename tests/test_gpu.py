@@ -109,6 +109,23 @@ def test_irregular_batch_sizes_eager_path(gpu_engine, packed):
         np.testing.assert_allclose(out["p_vals"], ref["p_vals"], atol=1e-6)
 
 
+def test_interleaved_batch_sizes(gpu_engine, packed):
+    """Batch sizes alternating large -> small -> large on one session.
+    Guards the accumulator all-zero invariant (finalize_kernel re-zeros the
+    rows it consumed instead of a per-request memset): a stale row from a
+    larger earlier batch would corrupt a later one."""
+    rng = np.random.default_rng(11)
+    for b in (4096, 64, 4096, 1, 2048, 300, 4096):
+        codes = np.stack(
+            [rng.integers(-1, len(v), size=b) for v in packed.vocabs], axis=1
+        ).astype(np.int16)
+        nums = rng.normal(5000.0, 3000.0, size=(b, 14)).astype(np.float32)
+        out = gpu_engine.score_arrays(codes, nums, with_drift=True)
+        ref = cpu_ref.score_batch_cpu(packed, codes, nums)
+        np.testing.assert_allclose(out["predictions"], ref["predictions"], atol=1e-9)
+        np.testing.assert_allclose(out["p_vals"], ref["p_vals"], atol=1e-6)
+
+
 def test_serving_stack_on_gpu(model_dir):
     """Whole serving stack on the GPU engine (TestClient): contract-valid
     responses through both the solo-flush bytes path and the merged path."""
