@@ -60,6 +60,7 @@ class ReplicaPool:
     def __init__(self, n: int, max_failures: int = 3):
         self.alive = [True] * n
         self.fails = [0] * n
+        self.max_failures = max_failures
         self._rr = count()
 
     def pick(self) -> int:
@@ -73,9 +74,9 @@ class ReplicaPool:
     def report_ok(self, i: int) -> None:
         self.fails[i] = 0
 
-    def report_fail(self, i: int, max_failures: int = 3) -> None:
+    def report_fail(self, i: int) -> None:
         self.fails[i] += 1
-        if self.fails[i] >= max_failures:
+        if self.fails[i] >= self.max_failures:
             self.alive[i] = False
 
 
