@@ -64,6 +64,8 @@ class MicroBatcher:
     async def submit(self, codes: np.ndarray, nums: np.ndarray) -> dict:
         """Submit one request's encoded rows; resolves to that request's slice
         of the flushed batch output."""
+        if self._closed:
+            raise RuntimeError("MicroBatcher is closed")
         fut = asyncio.get_running_loop().create_future()
         self._pending.append(_Pending(codes, nums, fut))
         self._pending_rows += len(codes)
