@@ -72,7 +72,7 @@ def main():
     p.add_argument("--model-trees", type=int, default=None,
                    help="override bench model n_estimators (sensitivity runs)")
     p.add_argument("--model-depth", type=int, default=None)
-    p.add_argument("--model-algo", default="rf", choices=["rf", "gbt"],
+    p.add_argument("--model-algo", default="rf", choices=["rf", "gbt", "et"],
                    help="classifier family for the bench model")
     args = p.parse_args()
     if args.model_trees:
@@ -288,7 +288,7 @@ def main():
                     "data": "synthetic (UCI-credit-default-shaped), random-seed-fitted model",
                     "config": {
                         "model": (
-                            f"{'GBT' if BENCH_MODEL.get('algo') == 'gbt' else 'RandomForest'} "
+                            f"{ {'gbt': 'GBT', 'et': 'ExtraTrees'}.get(BENCH_MODEL.get('algo'), 'RandomForest') } "
                             f"{BENCH_MODEL['n_estimators']}x"
                             f"depth{BENCH_MODEL['max_depth']} + IForest100 + TabularDrift"
                         ),

@@ -28,8 +28,9 @@ def _cmd_train(argv):
     p.add_argument("--min-roc-auc", type=float, default=None,
                    help="refuse to register below this validation ROC-AUC "
                         "(the quality gate the reference lacks)")
-    p.add_argument("--algorithm", default="rf", choices=["rf", "gbt"],
-                   help="rf = the reference's RandomForest; gbt = "
+    p.add_argument("--algorithm", default="rf", choices=["rf", "gbt", "et"],
+                   help="rf = the reference's RandomForest; et = ExtraTrees "
+                        "(same packed format/kernel); gbt = "
                         "gradient-boosted trees (same HIP traversal kernel)")
     p.add_argument("--data", default=None,
                    help="CSV with the UCI schema (the reference's curated "

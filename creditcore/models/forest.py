@@ -29,10 +29,12 @@ def make_classifier_pipeline(params: dict, algorithm: str = "rf") -> Pipeline:
     """Build the classifier pipeline (reference 01-train cell-6).
 
     ``params`` are estimator kwargs (n_estimators / max_depth / random_state
-    ...). ``algorithm``: "rf" (the reference's RandomForest) or "gbt"
+    ...). ``algorithm``: "rf" (the reference's RandomForest), "gbt"
     (GradientBoostingClassifier — the north star's "gradient-boosted-tree
     traversal" family; same packed node-SoA format, scored by the same HIP
-    kernel with a sigmoid(sum + prior) finalize).
+    kernel with a sigmoid(sum + prior) finalize), or "et"
+    (ExtraTreesClassifier — identical tree structure and leaf-fraction-mean
+    semantics to RF, so it rides the RF pack/score path unchanged).
     """
     if algorithm == "gbt":
         from sklearn.ensemble import GradientBoostingClassifier
@@ -41,6 +43,10 @@ def make_classifier_pipeline(params: dict, algorithm: str = "rf") -> Pipeline:
         params.pop("criterion", None)  # rf-only knob from the search space
         params.setdefault("n_estimators", 200)
         estimator = GradientBoostingClassifier(**params)
+    elif algorithm == "et":
+        from sklearn.ensemble import ExtraTreesClassifier
+
+        estimator = ExtraTreesClassifier(**params, n_jobs=-1)
     elif algorithm == "rf":
         estimator = RandomForestClassifier(**params, n_jobs=-1)
     else:
