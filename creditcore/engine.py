@@ -5,9 +5,10 @@ resident HBM model buffers, two private HIP streams, double-buffered pinned
 staging, and one hipGraph per recurring request shape. A request is:
 
     host:   native JSON parse -> codes (int16), nums (f32)   [C, GIL-free]
-    graph:  pinned H2D -> accumulator memset -> 2-tree-ILP forest kernels
-            (classifier + isolation forest) ∥ drift branch on stream 2
-            (categorical histogram + 512-thread K-S) -> finalize -> D2H
+    graph:  pinned H2D -> 2-tree-ILP forest kernels (classifier + isolation
+            forest) -> finalize (re-zeros its accumulator rows) -> outs D2H,
+            ∥ drift branch on stream 2 (categorical histogram + K-S) ->
+            one packed drift D2H; no memset nodes
     host:   drift p-values (C: chi2 closed forms + MTW/Pelz-Good K-S sf)
             -> response JSON bytes (C, shortest-round-trip doubles)
 
