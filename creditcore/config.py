@@ -47,6 +47,8 @@ class ServeConfig:
     # drift
     drift_sync_period: int = field(default_factory=lambda: _env("drift_sync_period", 64, int))
     drift_max_batch: int = field(default_factory=lambda: _env("drift_max_batch", 8192, int))
+    # non-empty => drift histograms persist across restarts at this path
+    drift_state_path: str = field(default_factory=lambda: _env("drift_state_path", ""))
 
     # observability
     log_inference_data: bool = field(default_factory=lambda: _env("log_inference_data", True, bool))

@@ -234,6 +234,48 @@ class DriftSync:
         replica's local histogram into this one (cross-GPU copy over xGMI)."""
         self.local += other.local.to(self.local.device)
 
+    # -------------------------------------------------- restart persistence
+    def save_state(self, path: str) -> None:
+        """Persist the local histogram so drift state survives a service
+        restart (the reference's pods lose all drift context on restart)."""
+        import os
+        import tempfile
+
+        d = os.path.dirname(os.path.abspath(path))
+        os.makedirs(d, exist_ok=True)
+        fd, tmp = tempfile.mkstemp(dir=d, suffix=".tmp")
+        try:
+            with os.fdopen(fd, "wb") as f:
+                np.savez(f, local=self.local.cpu().numpy(),
+                         batches=np.int64(self.batches), n_bins=np.int64(self.n_bins))
+            os.replace(tmp, path)
+        except BaseException:
+            try:
+                os.unlink(tmp)
+            except OSError:
+                pass
+            raise
+
+    def load_state(self, path: str) -> bool:
+        """Restore a saved histogram; returns False (and starts fresh) on a
+        missing file or a bin-layout mismatch (e.g. model/bins changed)."""
+        import os
+
+        import torch
+
+        if not os.path.isfile(path):
+            return False
+        try:
+            with np.load(path) as z:
+                if int(z["n_bins"]) != self.n_bins or z["local"].shape != (self.size,):
+                    return False
+                self.local = torch.from_numpy(z["local"].copy()).to(self.local.device)
+                self.batches = int(z["batches"])
+        except (OSError, ValueError, KeyError):
+            return False
+        self.allreduce()
+        return True
+
     # -------------------------------------------------- cross-process merge
     # SO_REUSEPORT serving workers are separate processes with no process
     # group; they publish local histograms to a shared directory (tmpfs)

@@ -48,6 +48,11 @@ class RawScoreServer:
         cfg = self.cfg
         self.engines = _build_engines(cfg)
         self.drift_sync = DriftSync(self.engines[0].packed, device="cpu")
+        if cfg.drift_state_path and cfg.workers <= 1:
+            # single-process serving only: SO_REUSEPORT workers already share
+            # live state via the tmpfs publish dir, and one shared file would
+            # lose all but the last worker's shard on shutdown
+            self.drift_sync.load_state(cfg.drift_state_path)
 
         def scorer(e):
             def run(codes, nums):
@@ -92,6 +97,8 @@ class RawScoreServer:
         await self._server.wait_closed()
         for b in self.batchers:
             await b.close()
+        if self.cfg.drift_state_path and self.cfg.workers <= 1:
+            self.drift_sync.save_state(self.cfg.drift_state_path)
 
     @property
     def _drift_dir(self) -> str:
@@ -180,7 +187,9 @@ class RawScoreServer:
         return 200, payload
 
     async def _get(self, path: bytes) -> tuple[int, bytes]:
-        path = path.partition(b"?")[0]
+        path, _, query = path.partition(b"?")
+        if path == b"/metrics" and b"format=prometheus" in query:
+            return 200, self.metrics.prometheus().encode(), b"text/plain; version=0.0.4"
         if path == b"/healthz":
             any_alive = any(self.pool.alive)
             return 200, json.dumps(
@@ -227,13 +236,17 @@ class RawScoreServer:
                         await self._respond(writer, 411, b'{"detail": "length required"}')
                         return
                 body = await reader.readexactly(clen) if clen else b""
+                ctype = b"application/json"
                 if method == b"POST" and path.partition(b"?")[0] in (b"/score", b"/predict"):
                     status, payload = await self._score(body)
                 elif method == b"GET":
-                    status, payload = await self._get(path)
+                    res = await self._get(path)
+                    status, payload = res[0], res[1]
+                    if len(res) > 2:
+                        ctype = res[2]
                 else:
                     status, payload = 405, b'{"detail": "method not allowed"}'
-                await self._respond(writer, status, payload)
+                await self._respond(writer, status, payload, ctype)
                 if not keep_alive:
                     break
         except (asyncio.IncompleteReadError, ConnectionResetError):
@@ -250,9 +263,10 @@ class RawScoreServer:
                 422: b"Unprocessable Entity", 500: b"Internal Server Error",
                 503: b"Service Unavailable"}
 
-    async def _respond(self, writer, status: int, payload: bytes):
-        head = b"HTTP/1.1 %d %s\r\nContent-Type: application/json\r\nContent-Length: %d\r\n\r\n" % (
-            status, self._REASONS.get(status, b"OK"), len(payload),
+    async def _respond(self, writer, status: int, payload: bytes,
+                       ctype: bytes = b"application/json"):
+        head = b"HTTP/1.1 %d %s\r\nContent-Type: %s\r\nContent-Length: %d\r\n\r\n" % (
+            status, self._REASONS.get(status, b"OK"), ctype, len(payload),
         )
         writer.write(head + payload)
         await writer.drain()
@@ -288,6 +302,9 @@ def _spawn_workers(cfg: ServeConfig):
     for f in dataclasses.fields(cfg):
         env_base[f"CREDITCORE_{f.name.upper()}"] = str(getattr(cfg, f.name))
     env_base["CREDITCORE_WORKERS"] = "1"  # workers serve single-process
+    # workers share live drift via the tmpfs publish dir; a shared state
+    # file would keep only the last worker's shard — disable per-worker
+    env_base["CREDITCORE_DRIFT_STATE_PATH"] = ""
     try:
         import torch
 

@@ -89,6 +89,8 @@ async def lifespan(app: FastAPI):
     # One shared node-level drift accumulator (host-side; 23×bins int64 —
     # the per-request drift stats feeding it come from the HIP kernels).
     drift_sync = DriftSync(engines[0].packed, device="cpu")
+    if cfg.drift_state_path:
+        drift_sync.load_state(cfg.drift_state_path)
 
     def _fold_drift(out, nums):
         if "cat_hist" in out:
@@ -145,6 +147,8 @@ async def lifespan(app: FastAPI):
     yield
     for b in batchers:
         await b.close()
+    if cfg.drift_state_path:
+        drift_sync.save_state(cfg.drift_state_path)
     state.clear()
 
 
@@ -313,7 +317,12 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         return body
 
     @app.get("/metrics")
-    async def metrics_endpoint():
+    async def metrics_endpoint(format: str = "json"):
+        if format == "prometheus":
+            return Response(
+                state["metrics"].prometheus(),
+                media_type="text/plain; version=0.0.4",
+            )
         return state["metrics"].snapshot()
 
     @app.post("/predict_dense")

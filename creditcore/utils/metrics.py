@@ -52,3 +52,25 @@ class Metrics:
                 "latency_ms_p90": pct(0.90),
                 "latency_ms_p99": pct(0.99),
             }
+
+    def prometheus(self, prefix: str = "creditcore") -> str:
+        """Render the snapshot in the Prometheus text exposition format
+        (what a K8s scrape expects — the reference shipped logs to Log
+        Analytics instead and had no scrape endpoint)."""
+        s = self.snapshot()
+        lines = []
+        for name, kind in (
+            ("requests_total", "counter"),
+            ("rows_total", "counter"),
+            ("errors_total", "counter"),
+            ("drift_syncs_total", "counter"),
+            ("uptime_s", "gauge"),
+        ):
+            lines.append(f"# TYPE {prefix}_{name} {kind}")
+            lines.append(f"{prefix}_{name} {s[name]}")
+        lines.append(f"# TYPE {prefix}_latency_ms summary")
+        for q, key in (("0.5", "latency_ms_p50"), ("0.9", "latency_ms_p90"),
+                       ("0.99", "latency_ms_p99")):
+            if s[key] is not None:
+                lines.append(f'{prefix}_latency_ms{{quantile="{q}"}} {s[key]}')
+        return "\n".join(lines) + "\n"

@@ -1,0 +1,106 @@
+"""Ops surface: Prometheus-format /metrics and drift-state persistence
+across service restarts (both absent from the reference, which shipped logs
+to Log Analytics and lost all drift context on pod restart)."""
+
+from __future__ import annotations
+
+import numpy as np
+import pytest
+import torch
+
+from creditcore.utils.metrics import Metrics
+
+
+def test_prometheus_rendering():
+    m = Metrics()
+    m.observe_request(64, 1.5)
+    m.observe_request(64, 2.5)
+    m.observe_error()
+    text = m.prometheus()
+    assert "# TYPE creditcore_requests_total counter" in text
+    assert "creditcore_requests_total 2" in text
+    assert "creditcore_rows_total 128" in text
+    assert "creditcore_errors_total 1" in text
+    assert 'creditcore_latency_ms{quantile="0.5"}' in text
+    assert text.endswith("\n")
+
+
+def test_metrics_endpoint_prometheus(model_dir):
+    from fastapi.testclient import TestClient
+
+    from creditcore.config import ServeConfig
+    from creditcore.schema import SAMPLE_REQUEST
+    from creditcore.serve import create_app
+
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    with TestClient(create_app(cfg)) as client:
+        assert client.post("/score", json=SAMPLE_REQUEST).status_code == 200
+        r = client.get("/metrics?format=prometheus")
+        assert r.status_code == 200
+        assert r.headers["content-type"].startswith("text/plain")
+        assert "creditcore_requests_total 1" in r.text
+        # default stays JSON
+        assert client.get("/metrics").json()["requests_total"] == 1
+
+
+def _fold_batch(ds, packed, seed):
+    from creditcore.data import make_request_batch
+    from creditcore.ops import cpu_ref
+    from creditcore.pack import encode_batch
+
+    recs = make_request_batch(90, seed=seed)
+    codes, nums = encode_batch(recs, packed.vocabs)
+    nums_imp = cpu_ref.impute_nums(packed, nums)
+    hist, _ = cpu_ref.drift_stats_cpu(packed, codes, nums_imp)
+    ds.accumulate(torch.from_numpy(hist), torch.from_numpy(nums))
+
+
+def test_driftsync_state_roundtrip(packed, tmp_path):
+    from creditcore.parallel import DriftSync
+
+    p = str(tmp_path / "state" / "drift.npz")
+    a = DriftSync(packed, device="cpu", n_bins=16)
+    _fold_batch(a, packed, 1)
+    _fold_batch(a, packed, 2)
+    a.save_state(p)
+
+    b = DriftSync(packed, device="cpu", n_bins=16)
+    assert b.load_state(p)
+    np.testing.assert_array_equal(a.local.numpy(), b.local.numpy())
+    assert b.batches == 2
+    b.allreduce()
+    assert b.snapshot()["rows"] == 180
+
+    # layout mismatch (different binning) must refuse and start fresh
+    c = DriftSync(packed, device="cpu", n_bins=8)
+    assert not c.load_state(p)
+    assert int(c.local.sum()) == 0
+    # missing file
+    assert not c.load_state(str(tmp_path / "nope.npz"))
+
+
+def test_drift_state_survives_service_restart(model_dir, tmp_path):
+    from fastapi.testclient import TestClient
+
+    from creditcore.config import ServeConfig
+    from creditcore.data import make_request_batch
+    from creditcore.serve import create_app
+
+    path = str(tmp_path / "drift_state.npz")
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    cfg.drift_state_path = path
+
+    with TestClient(create_app(cfg)) as client:
+        assert client.post("/score", json=make_request_batch(50, seed=3)).status_code == 200
+        rows_before = client.get("/drift").json()["rows"]
+    assert rows_before == 50
+
+    # fresh app instance = restarted service: state is restored from disk
+    with TestClient(create_app(cfg)) as client:
+        assert client.get("/drift").json()["rows"] == 50
+        client.post("/score", json=make_request_batch(25, seed=4))
+        assert client.get("/drift").json()["rows"] == 75
