@@ -80,24 +80,18 @@ class ReplicaPool:
             self.alive[i] = False
 
 
-@asynccontextmanager
-async def lifespan(app: FastAPI):
-    from .parallel import DriftSync
-
-    cfg: ServeConfig = app.state.cfg
-    engines = _build_engines(cfg)
-    # One shared node-level drift accumulator (host-side; 23×bins int64 —
-    # the per-request drift stats feeding it come from the HIP kernels).
-    drift_sync = DriftSync(engines[0].packed, device="cpu")
-    if cfg.drift_state_path:
-        drift_sync.load_state(cfg.drift_state_path)
+def _make_batchers(cfg: ServeConfig, engines, drift_sync) -> list[MicroBatcher]:
+    """One micro-batcher per engine, each folding its drift stats into the
+    shared node-level accumulator (allreduce every drift_sync_period)."""
 
     def _fold_drift(out, nums):
         if "cat_hist" in out:
             drift_sync.accumulate(out["cat_hist"], nums)
             if drift_sync.batches % max(cfg.drift_sync_period, 1) == 0:
                 drift_sync.allreduce()
-                state["metrics"].observe_drift_sync()
+                m = state.get("metrics")
+                if m is not None:
+                    m.observe_drift_sync()
 
     def scorer(e: ScoringEngine):
         def run(codes, nums):
@@ -119,7 +113,7 @@ async def lifespan(app: FastAPI):
 
         return run
 
-    batchers = [
+    return [
         MicroBatcher(
             scorer(e),
             max_rows=cfg.max_batch_rows,
@@ -128,6 +122,21 @@ async def lifespan(app: FastAPI):
         )
         for e in engines
     ]
+
+
+@asynccontextmanager
+async def lifespan(app: FastAPI):
+    from .parallel import DriftSync
+
+    cfg: ServeConfig = app.state.cfg
+    engines = _build_engines(cfg)
+    # One shared node-level drift accumulator (host-side; 23×bins int64 —
+    # the per-request drift stats feeding it come from the HIP kernels).
+    drift_sync = DriftSync(engines[0].packed, device="cpu")
+    if cfg.drift_state_path:
+        drift_sync.load_state(cfg.drift_state_path)
+
+    batchers = _make_batchers(cfg, engines, drift_sync)
     for b in batchers:
         await b.start()
     dense_engine = None
@@ -145,10 +154,11 @@ async def lifespan(app: FastAPI):
     state["metrics"] = Metrics()
     state["cfg"] = cfg
     yield
-    for b in batchers:
+    # use the *current* objects — /admin/reload may have swapped them
+    for b in state.get("batchers", batchers):
         await b.close()
     if cfg.drift_state_path:
-        drift_sync.save_state(cfg.drift_state_path)
+        state.get("drift_sync", drift_sync).save_state(cfg.drift_state_path)
     state.clear()
 
 
@@ -365,6 +375,45 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
                 ).astype(np.float64).tolist(),
             }
         )
+
+    @app.post("/admin/reload")
+    async def reload_model(request: Request):
+        """Hot-swap the served model without a restart — the in-process
+        equivalent of the reference's rolling redeploy (its only path to a
+        new model version was rebuilding + redeploying every container).
+        Body: {"model_uri": "models:/<name>/<version|latest>" | <dir>}.
+        Builds the new engines first, then swaps atomically; in-flight
+        requests finish on the old engines."""
+        from .parallel import DriftSync
+
+        cfg: ServeConfig = state["cfg"]
+        try:
+            payload = json.loads(await request.body() or b"{}")
+        except ValueError:
+            raise HTTPException(status_code=422, detail="body must be JSON")
+        uri = payload.get("model_uri") or cfg.model_directory
+        import copy
+        import dataclasses
+
+        new_cfg = dataclasses.replace(copy.copy(cfg), model_directory=uri)
+        try:
+            engines = _build_engines(new_cfg)
+        except Exception as e:
+            raise HTTPException(status_code=422, detail=f"cannot load {uri!r}: {e}")
+
+        drift_sync = DriftSync(engines[0].packed, device="cpu")
+        batchers = _make_batchers(new_cfg, engines, drift_sync)
+        for b in batchers:
+            await b.start()
+        old_batchers = state["batchers"]
+        state["engines"] = engines
+        state["batchers"] = batchers
+        state["pool"] = ReplicaPool(len(engines))
+        state["drift_sync"] = drift_sync  # new model ⇒ new drift reference
+        state["cfg"] = new_cfg
+        for b in old_batchers:  # drain + stop after the swap
+            await b.close()
+        return {"status": "reloaded", "model_uri": uri, "engines": len(engines)}
 
     @app.get("/drift")
     async def drift_endpoint():

@@ -104,3 +104,47 @@ def test_drift_state_survives_service_restart(model_dir, tmp_path):
         assert client.get("/drift").json()["rows"] == 50
         client.post("/score", json=make_request_batch(25, seed=4))
         assert client.get("/drift").json()["rows"] == 75
+
+
+def test_hot_reload_swaps_model(model_dir, train_df, tmp_path):
+    """/admin/reload swaps to a new model version without a restart (the
+    reference's only path to a new version was a full redeploy)."""
+    from fastapi.testclient import TestClient
+
+    from creditcore import registry
+    from creditcore.config import ServeConfig
+    from creditcore.data import make_request_batch
+    from creditcore.models.forest import make_classifier_pipeline
+    from creditcore.schema import FEATURES, TARGET
+    from creditcore.serve import create_app
+    from creditcore.train import fit_detectors
+
+    # a deliberately different second model (tiny, different seed/shape)
+    pipe = make_classifier_pipeline(
+        {"n_estimators": 7, "max_depth": 2, "random_state": 99}
+    )
+    sub = train_df.head(1200)
+    pipe.fit(sub[FEATURES], sub[TARGET].values.ravel())
+    drift, outlier = fit_detectors(sub)
+    v2_dir = str(tmp_path / "v2")
+    registry.save_pyfunc_model(v2_dir, pipe, drift, outlier)
+
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    batch = make_request_batch(64, seed=5)
+    with TestClient(create_app(cfg)) as client:
+        before = client.post("/score", json=batch).json()["predictions"]
+        r = client.post("/admin/reload", json={"model_uri": v2_dir})
+        assert r.status_code == 200
+        assert r.json()["status"] == "reloaded"
+        after = client.post("/score", json=batch).json()["predictions"]
+        # different model -> different scores; service never restarted
+        assert before != after
+        assert client.get("/healthz").json()["status"] == "ok"
+        # bad URI is rejected and the live model keeps serving
+        assert client.post(
+            "/admin/reload", json={"model_uri": str(tmp_path / "missing")}
+        ).status_code == 422
+        again = client.post("/score", json=batch).json()["predictions"]
+        assert again == after
