@@ -43,6 +43,8 @@ class ServeConfig:
     # micro-batching
     max_batch_rows: int = field(default_factory=lambda: _env("max_batch_rows", 8192, int))
     batch_wait_us: int = field(default_factory=lambda: _env("batch_wait_us", 300, int))
+    # request-body cap for the raw frontend (413 above this)
+    max_body_bytes: int = field(default_factory=lambda: _env("max_body_bytes", 64 << 20, int))
 
     # drift
     drift_sync_period: int = field(default_factory=lambda: _env("drift_sync_period", 64, int))

@@ -235,6 +235,11 @@ class RawScoreServer:
                         # chunked bodies: not needed by the contract clients
                         await self._respond(writer, 411, b'{"detail": "length required"}')
                         return
+                if clen > self.cfg.max_body_bytes:
+                    # refuse before reading: an unbounded Content-Length must
+                    # not drive readexactly into allocating it
+                    await self._respond(writer, 413, b'{"detail": "body too large"}')
+                    return
                 body = await reader.readexactly(clen) if clen else b""
                 ctype = b"application/json"
                 if method == b"POST" and path.partition(b"?")[0] in (b"/score", b"/predict"):
@@ -260,8 +265,8 @@ class RawScoreServer:
 
     _REASONS = {200: b"OK", 400: b"Bad Request", 404: b"Not Found",
                 405: b"Method Not Allowed", 411: b"Length Required",
-                422: b"Unprocessable Entity", 500: b"Internal Server Error",
-                503: b"Service Unavailable"}
+                413: b"Payload Too Large", 422: b"Unprocessable Entity",
+                500: b"Internal Server Error", 503: b"Service Unavailable"}
 
     async def _respond(self, writer, status: int, payload: bytes,
                        ctype: bytes = b"application/json"):

@@ -98,6 +98,21 @@ def test_metrics_and_drift(raw_url):
     assert len(d["node_feature_drift"]) == 23
 
 
+def test_oversized_body_gets_413(raw_url):
+    """A hostile Content-Length must be refused before the read, not
+    buffered into memory."""
+    import socket
+
+    host, port = raw_url[len("http://"):].rsplit(":", 1)
+    with socket.create_connection((host, int(port)), timeout=15) as s:
+        s.sendall(
+            b"POST /score HTTP/1.1\r\nHost: x\r\n"
+            b"Content-Length: 99999999999\r\n\r\n"
+        )
+        data = s.recv(4096)
+    assert data.startswith(b"HTTP/1.1 413")
+
+
 def test_chunked_body_gets_411(raw_url):
     """Transfer-Encoding: chunked is outside the contract clients' subset —
     the frontend must refuse it deterministically, not misparse it."""
