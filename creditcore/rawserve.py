@@ -278,11 +278,29 @@ class RawScoreServer:
 
 
 async def serve_raw(cfg: ServeConfig):
+    import signal as _signal
+
     server = RawScoreServer(cfg)
     await server.start()
     print(f"[rawserve] listening on {cfg.host}:{cfg.port}", flush=True)
-    async with server._server:
-        await server._server.serve_forever()
+    # graceful SIGTERM/SIGINT (K8s pod stop): close the listener, then run
+    # the shutdown hooks — drift-state persistence happens in close()
+    stop = asyncio.Event()
+    loop = asyncio.get_running_loop()
+    for sig in (_signal.SIGTERM, _signal.SIGINT):
+        try:
+            loop.add_signal_handler(sig, stop.set)
+        except (ValueError, NotImplementedError):  # non-main thread
+            pass
+    serve_task = asyncio.create_task(server._server.serve_forever())
+    await stop.wait()
+    serve_task.cancel()
+    try:
+        await serve_task
+    except asyncio.CancelledError:
+        pass
+    await server.close()
+    print("[rawserve] shut down cleanly", flush=True)
 
 
 def main(cfg: ServeConfig):

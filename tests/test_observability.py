@@ -148,3 +148,63 @@ def test_hot_reload_swaps_model(model_dir, train_df, tmp_path):
         ).status_code == 422
         again = client.post("/score", json=batch).json()["predictions"]
         assert again == after
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(120)
+def test_raw_server_sigterm_persists_drift(model_dir, tmp_path):
+    """SIGTERM (the K8s pod-stop signal) must shut the raw server down
+    through close(), persisting drift state."""
+    import os
+    import signal
+    import socket
+    import subprocess
+    import sys
+    import time
+
+    import httpx
+
+    import creditcore
+    from creditcore.data import make_request_batch
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    state_path = str(tmp_path / "drift.npz")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(creditcore.__file__)))
+    env = dict(os.environ, CREDITCORE_LOG_INFERENCE_DATA="0",
+               CREDITCORE_LOG_RESPONSES="0",
+               PYTHONPATH=repo + os.pathsep + os.environ.get("PYTHONPATH", ""))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "creditcore", "serve", "--raw-http",
+         "--device", "cpu", "--host", "127.0.0.1", "--port", str(port),
+         "--model-directory", model_dir, "--drift-state-path", state_path],
+        env=env, cwd=str(tmp_path),
+    )
+    try:
+        deadline = time.monotonic() + 90
+        ok = False
+        while time.monotonic() < deadline and not ok:
+            try:
+                r = httpx.post(f"http://127.0.0.1:{port}/score",
+                               json=make_request_batch(40, seed=9), timeout=10.0)
+                ok = r.status_code == 200
+            except Exception:
+                time.sleep(0.5)
+        assert ok
+        proc.send_signal(signal.SIGTERM)
+        assert proc.wait(timeout=30) == 0
+        assert os.path.isfile(state_path)
+
+        from creditcore.pack import pack_pyfunc_dir
+        from creditcore.parallel import DriftSync
+
+        ds = DriftSync(pack_pyfunc_dir(model_dir), device="cpu")
+        assert ds.load_state(state_path)
+        assert ds.snapshot()["rows"] == 40
+    finally:
+        try:
+            proc.kill()
+        except Exception:
+            pass
