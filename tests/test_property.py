@@ -117,3 +117,30 @@ def test_encoder_rejects_junk_bytes(packed, junk):
         assert codes.shape[1] == 9
     except ValueError:
         pass
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    seed=st.integers(0, 10_000),
+    rows=st.integers(1, 200),
+    nan_frac=st.floats(0.0, 0.4),
+)
+def test_driftsync_torch_and_numpy_paths_agree(packed, seed, rows, nan_frac):
+    """DriftSync.accumulate has a torch path (GPU tensors) and a numpy fast
+    path — they must produce identical histograms, NaNs included."""
+    import torch
+
+    from creditcore.parallel import DriftSync
+
+    rng = np.random.default_rng(seed)
+    C = int(packed.ref_cat_offsets[-1])
+    cat_hist = rng.integers(0, 50, size=C).astype(np.int32)
+    nums = rng.normal(5000.0, 4000.0, size=(rows, 14)).astype(np.float32)
+    nums[rng.uniform(size=nums.shape) < nan_frac] = np.nan
+
+    a = DriftSync(packed, device="cpu", n_bins=16)
+    b = DriftSync(packed, device="cpu", n_bins=16)
+    a.accumulate(torch.from_numpy(cat_hist.copy()), torch.from_numpy(nums.copy()))
+    b.accumulate(cat_hist, nums)  # numpy fast path
+    np.testing.assert_array_equal(a.local.numpy(), b.local.numpy())
+    assert a.batches == b.batches == 1
