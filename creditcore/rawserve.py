@@ -40,19 +40,8 @@ class RawScoreServer:
         self.drift_sync = None
         self.pool = None
 
-    # ------------------------------------------------------------ lifecycle
-    async def start(self):
-        from .parallel import DriftSync
-        from .serve import _build_engines
-
+    def _wire_batchers(self, engines) -> list[MicroBatcher]:
         cfg = self.cfg
-        self.engines = _build_engines(cfg)
-        self.drift_sync = DriftSync(self.engines[0].packed, device="cpu")
-        if cfg.drift_state_path and cfg.workers <= 1:
-            # single-process serving only: SO_REUSEPORT workers already share
-            # live state via the tmpfs publish dir, and one shared file would
-            # lose all but the last worker's shard on shutdown
-            self.drift_sync.load_state(cfg.drift_state_path)
 
         def scorer(e):
             def run(codes, nums):
@@ -73,15 +62,31 @@ class RawScoreServer:
 
             return run
 
-        self.batchers = [
+        return [
             MicroBatcher(
                 scorer(e),
                 max_rows=cfg.max_batch_rows,
                 max_wait_us=cfg.batch_wait_us,
                 score_single=scorer_single(e),
             )
-            for e in self.engines
+            for e in engines
         ]
+
+    # ------------------------------------------------------------ lifecycle
+    async def start(self):
+        from .parallel import DriftSync
+        from .serve import _build_engines
+
+        cfg = self.cfg
+        self.engines = _build_engines(cfg)
+        self.drift_sync = DriftSync(self.engines[0].packed, device="cpu")
+        if cfg.drift_state_path and cfg.workers <= 1:
+            # single-process serving only: SO_REUSEPORT workers already share
+            # live state via the tmpfs publish dir, and one shared file would
+            # lose all but the last worker's shard on shutdown
+            self.drift_sync.load_state(cfg.drift_state_path)
+
+        self.batchers = self._wire_batchers(self.engines)
         for b in self.batchers:
             await b.start()
         self.pool = ReplicaPool(len(self.engines))
@@ -186,6 +191,41 @@ class RawScoreServer:
             )
         return 200, payload
 
+    async def _reload(self, body: bytes) -> tuple[int, bytes]:
+        """Hot model swap (per worker): build new engines first, then swap
+        atomically; in-flight requests finish on the old ones. Body:
+        {"model_uri": "models:/<name>/<version|latest>" | <dir>}."""
+        import dataclasses
+
+        from .parallel import DriftSync
+        from .serve import _build_engines
+
+        try:
+            payload = json.loads(body or b"{}")
+        except ValueError:
+            return 422, b'{"detail": "body must be JSON"}'
+        uri = payload.get("model_uri") or self.cfg.model_directory
+        new_cfg = dataclasses.replace(self.cfg, model_directory=uri)
+        try:
+            engines = _build_engines(new_cfg)
+        except Exception as e:
+            return 422, json.dumps({"detail": f"cannot load {uri!r}: {e}"}).encode()
+        drift_sync = DriftSync(engines[0].packed, device="cpu")
+        batchers = self._wire_batchers(engines)
+        for b in batchers:
+            await b.start()
+        old = self.batchers
+        self.engines = engines
+        self.batchers = batchers
+        self.pool = ReplicaPool(len(engines))
+        self.drift_sync = drift_sync  # new model => new drift reference
+        self.cfg = new_cfg
+        for b in old:
+            await b.close()
+        return 200, json.dumps(
+            {"status": "reloaded", "model_uri": uri, "engines": len(engines)}
+        ).encode()
+
     async def _get(self, path: bytes) -> tuple[int, bytes]:
         path, _, query = path.partition(b"?")
         if path == b"/metrics" and b"format=prometheus" in query:
@@ -242,8 +282,11 @@ class RawScoreServer:
                     return
                 body = await reader.readexactly(clen) if clen else b""
                 ctype = b"application/json"
-                if method == b"POST" and path.partition(b"?")[0] in (b"/score", b"/predict"):
+                ppath = path.partition(b"?")[0]
+                if method == b"POST" and ppath in (b"/score", b"/predict"):
                     status, payload = await self._score(body)
+                elif method == b"POST" and ppath == b"/admin/reload":
+                    status, payload = await self._reload(body)
                 elif method == b"GET":
                     res = await self._get(path)
                     status, payload = res[0], res[1]

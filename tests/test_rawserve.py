@@ -98,6 +98,26 @@ def test_metrics_and_drift(raw_url):
     assert len(d["node_feature_drift"]) == 23
 
 
+def test_admin_reload_on_raw_frontend(raw_url, model_dir):
+    """Hot swap works on the raw frontend too; an identity reload (same
+    model dir) must leave scoring behavior unchanged, and a bad URI must be
+    a 422 with the live model still serving."""
+    import httpx
+
+    from creditcore.schema import SAMPLE_REQUEST
+
+    before = httpx.post(f"{raw_url}/score", json=SAMPLE_REQUEST, timeout=60.0).json()
+    r = httpx.post(f"{raw_url}/admin/reload", json={"model_uri": model_dir},
+                   timeout=120.0)
+    assert r.status_code == 200 and r.json()["status"] == "reloaded"
+    after = httpx.post(f"{raw_url}/score", json=SAMPLE_REQUEST, timeout=60.0).json()
+    assert after["predictions"] == before["predictions"]
+    assert httpx.post(f"{raw_url}/admin/reload", json={"model_uri": "/nope"},
+                      timeout=60.0).status_code == 422
+    still = httpx.post(f"{raw_url}/score", json=SAMPLE_REQUEST, timeout=60.0)
+    assert still.status_code == 200
+
+
 def test_oversized_body_gets_413(raw_url):
     """A hostile Content-Length must be refused before the read, not
     buffered into memory."""
