@@ -1,0 +1,86 @@
+"""bench.py driver-contract rehearsal on CPU: the exact invocation shapes
+the round-end driver uses (single-process, and torchrun ws=2 over gloo),
+checked for the one-JSON-line output contract."""
+
+from __future__ import annotations
+
+import json
+import os
+import socket
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+REQUIRED_KEYS = {
+    "metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+    "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config",
+}
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _last_json_line(stdout: str) -> dict:
+    lines = [l for l in stdout.strip().splitlines() if l.startswith("{")]
+    assert lines, f"no JSON line in output:\n{stdout[-2000:]}"
+    return json.loads(lines[-1])
+
+
+def _env(tmp_path):
+    return dict(
+        os.environ,
+        PYTHONPATH=REPO + os.pathsep + os.environ.get("PYTHONPATH", ""),
+        TMPDIR=str(tmp_path),  # isolate the packed-model cache
+    )
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(600)
+def test_bench_single_process_contract(tmp_path):
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--device", "cpu",
+         "--steps", "2", "--warmup", "1", "--rows", "64",
+         "--model-trees", "12", "--model-depth", "4"],
+        capture_output=True, text=True, timeout=540, env=_env(tmp_path), cwd=REPO,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    d = _last_json_line(r.stdout)
+    assert REQUIRED_KEYS <= set(d)
+    assert d["n_gpus"] == 1 and d["steps"] == 2 and d["warmup"] == 1
+    assert d["higher_is_better"] is True and d["scaling"] == "weak"
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["config"]["parallelism"] == "dp1"
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(600)
+def test_bench_torchrun_ws2_gloo(tmp_path):
+    """The driver's N>1 launch shape (torch.distributed.run, one rank per
+    GPU) rehearsed with gloo on CPU: rank 0 trains + broadcasts, both ranks
+    step, MAX-over-ranks elapsed, rank 0 prints one line with the
+    whole-job aggregate."""
+    port = _free_port()
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         os.path.join(REPO, "bench.py"), "--device", "cpu", "--gpus", "2",
+         "--steps", "2", "--warmup", "1", "--rows", "64",
+         "--model-trees", "12", "--model-depth", "4"],
+        capture_output=True, text=True, timeout=540, env=_env(tmp_path), cwd=REPO,
+    )
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    d = _last_json_line(r.stdout)
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 2 * 64
+    # exactly one result line (rank 0 only)
+    assert sum(1 for l in r.stdout.splitlines() if l.startswith("{")) == 1
