@@ -1057,8 +1057,13 @@ struct ScoreSession {
         (size_t)b * N_CAT * sizeof(short), hipMemcpyHostToDevice, stream));
     HIP_CHECK(hipMemcpyAsync(d_nums.data_ptr(), p_nums(slot),
         (size_t)b * N_NUM * sizeof(float), hipMemcpyHostToDevice, stream));
-    // fork point: the drift branch (stream2) depends only on the H2D copies
-    if (with_drift) {
+    // fork point: the drift branch (stream2) depends only on the H2D
+    // copies. At tiny batches the drift kernels are microseconds and the
+    // fork/join event edges cost more graph-replay overhead than the
+    // overlap saves — run the branch inline on `stream` instead.
+    const bool fork_drift = with_drift && b > 64;
+    hipStream_t sdrift = fork_drift ? stream2 : stream;
+    if (fork_drift) {
       HIP_CHECK(hipEventRecord(ev_fork, stream));
       HIP_CHECK(hipStreamWaitEvent(stream2, ev_fork, 0));
     }
@@ -1091,16 +1096,17 @@ struct ScoreSession {
         proba, proba + b, proba + 2 * b);
 
     if (with_drift) {
-      // Drift branch forked onto stream2 right after the H2D copies: the
-      // K-S sort+scan and categorical histogram overlap the forest kernels
-      // (captured as parallel graph branches; joined before the end).
+      // Drift branch: when forked (b > 64) the K-S and categorical
+      // histogram run on stream2 in parallel with the forest chain
+      // (captured as parallel graph branches, joined after the output
+      // copy); at tiny batches everything stays serial on `stream`.
       // small batches: one block overwrites the histogram (no memset node);
       // larger ones pre-zero + atomically accumulate across blocks
       const int hist_blocks = (b <= 2048) ? 1 : std::min(row_blocks, 1024);
       if (hist_blocks > 1)
-        HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), stream2));
+        HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), sdrift));
       hipLaunchKernelGGL(cat_hist_kernel, dim3(hist_blocks), dim3(BLOCK),
-          (size_t)total_bins * sizeof(int), stream2,
+          (size_t)total_bins * sizeof(int), sdrift,
           d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
           d_hist());
       if (b <= KS_COUNT_MAX_ROWS) {
@@ -1108,7 +1114,7 @@ struct ScoreSession {
         // digit us at b=1024; see profiles/kernel_tuning.md)
         const size_t smem = (size_t)std::max(b, 512 / 64) * sizeof(float);
         hipLaunchKernelGGL((ks_count_kernel_t<512>), dim3(N_NUM), dim3(512),
-            smem, stream2,
+            smem, sdrift,
             d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b,
             ref_sorted.data_ptr<float>(), rs_off.data_ptr<int64_t>(), d_ksd());
       } else {
@@ -1120,22 +1126,22 @@ struct ScoreSession {
         // 512 threads: halves the bitonic's serial depth per thread
         // (kernel_micro: 48.6->31.2 us @1k, 963->509 @16k)
         hipLaunchKernelGGL((ks_kernel_t<512>), dim3(N_NUM), dim3(512),
-            (size_t)m_pow2 * sizeof(float), stream2,
+            (size_t)m_pow2 * sizeof(float), sdrift,
             d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b,
             m_pow2, /*ref_lds=*/0, ref_sorted.data_ptr<float>(),
             rs_off.data_ptr<int64_t>(), d_ksd());
       }
       // one D2H for the whole drift branch (hist + K-S D share a blob)
       HIP_CHECK(hipMemcpyAsync(p_drift(slot), d_drift.data_ptr<uint8_t>(),
-          drift_bytes, hipMemcpyDeviceToHost, stream2));
-      HIP_CHECK(hipEventRecord(ev_join, stream2));
+          drift_bytes, hipMemcpyDeviceToHost, sdrift));
+      if (fork_drift) HIP_CHECK(hipEventRecord(ev_join, sdrift));
     }
     // classifier-output D2H depends only on finalize — it overlaps the
     // drift branch's K-S tail; the join lands after it so graph completion
     // still covers both streams
     HIP_CHECK(hipMemcpyAsync(p_outs(slot), proba,
         (size_t)(3 * b) * sizeof(double), hipMemcpyDeviceToHost, stream));
-    if (with_drift) HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
+    if (fork_drift) HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
     HIP_CHECK(hipGetLastError());
   }
 
