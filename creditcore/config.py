@@ -48,13 +48,23 @@ class ServeConfig:
 
     # drift
     drift_sync_period: int = field(default_factory=lambda: _env("drift_sync_period", 64, int))
-    drift_max_batch: int = field(default_factory=lambda: _env("drift_max_batch", 8192, int))
+    # drift-sample cap per request batch (rows beyond it are still scored;
+    # drift statistics use the first drift_max_batch rows). Hardware ceiling
+    # is 16384 (the K-S kernel's LDS sort capacity); values above clamp.
+    drift_max_batch: int = field(default_factory=lambda: _env("drift_max_batch", 16384, int))
     # non-empty => drift histograms persist across restarts at this path
     drift_state_path: str = field(default_factory=lambda: _env("drift_state_path", ""))
 
     # observability
     log_inference_data: bool = field(default_factory=lambda: _env("log_inference_data", True, bool))
     log_responses: bool = field(default_factory=lambda: _env("log_responses", True, bool))
+
+    # admin endpoints (/admin/reload). The reload body names an arbitrary
+    # on-disk path that gets deserialized, so it must not be reachable by
+    # any client of the public listener: with no token set, only loopback
+    # clients may call it; with a token set, any client presenting it
+    # (Authorization: Bearer <t> or X-Admin-Token: <t>) may.
+    admin_token: str = field(default_factory=lambda: _env("admin_token", ""))
 
     @classmethod
     def from_args(cls, argv: list[str] | None = None) -> "ServeConfig":

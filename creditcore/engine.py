@@ -35,17 +35,29 @@ from .schema import FEATURES
 
 
 class ScoringEngine:
-    def __init__(self, packed: PackedModel, device: str = "cpu", device_index: int = 0):
+    # K-S kernel LDS sort capacity (csrc MAX_DRIFT_ROWS) — the hardware
+    # ceiling; the per-engine cap (config drift_max_batch) can only lower it
+    HW_DRIFT_MAX_ROWS = 16384
+
+    def __init__(
+        self,
+        packed: PackedModel,
+        device: str = "cpu",
+        device_index: int = 0,
+        drift_max_rows: int | None = None,
+    ):
         self.packed = packed
         self.device = device
         self.device_index = device_index
         self.n_features = N_CAT + N_NUM
+        self.DRIFT_MAX_ROWS = min(
+            drift_max_rows or self.HW_DRIFT_MAX_ROWS, self.HW_DRIFT_MAX_ROWS
+        )
         self._gpu = None
         if device == "cuda":
             self._init_gpu()
 
     # ------------------------------------------------------------------ GPU
-    DRIFT_MAX_ROWS = 16384  # K-S kernel LDS sort capacity (csrc MAX_DRIFT_ROWS)
 
     def _init_gpu(self, capacity: int = 16384):
         import torch
@@ -150,7 +162,18 @@ class ScoringEngine:
             return np.asarray(codes), np.asarray(nums)
         import json
 
-        return encode_batch(json.loads(body), self.packed.vocabs)
+        recs = json.loads(body)
+        if not isinstance(recs, list) or not all(isinstance(r, dict) for r in recs):
+            raise ValueError("body must be a JSON array of records")
+        # absent fields take the schema defaults — pydantic default
+        # semantics (reference app/model.py:8-34), matching the native
+        # parser's default-row fill (encode_batch alone would fill missing
+        # categoricals with MISSING_CATEGORY instead)
+        from .schema import LoanApplicant
+
+        defaults = LoanApplicant().__dict__
+        recs = [{**defaults, **r} for r in recs]
+        return encode_batch(recs, self.packed.vocabs)
 
     def _ensure_json_encoder(self):
         enc = getattr(self, "_json_encoder", None)
@@ -191,6 +214,7 @@ class ScoringEngine:
                 self.packed.ref_cat_offsets,
                 int(self.packed.ref_sorted_offsets[1] - self.packed.ref_sorted_offsets[0]),
                 FEATURES,
+                self.DRIFT_MAX_ROWS,
             )
         except RuntimeError:
             # batch larger than the resident session capacity: grow + retry
@@ -235,13 +259,21 @@ class ScoringEngine:
             }
         g = self._gpu
         self._ensure_capacity(b)
-        drift_now = b <= self.DRIFT_MAX_ROWS
         g["np_codes"][0][:b] = codes
         g["np_nums"][0][:b] = nums
-        g["sess"].score(b, drift_now, True)
         nb = b
-        if not drift_now:
+        if b <= self.DRIFT_MAX_ROWS:
+            g["sess"].score(b, True, True)
+        else:
+            # Oversized batch: drift is a batch-population statistic, so run
+            # the capped-sample drift pass FIRST (only with_drift passes
+            # write the pinned drift blob), then the full batch without
+            # drift — pin_outs then holds the b-packed layout
+            # build_response_json reads. The reverse order overwrote the
+            # b-packed outputs with a cap-packed second pass and corrupted
+            # rows >= cap (round-1 advisor finding).
             g["sess"].score(self.DRIFT_MAX_ROWS, True, True)
+            g["sess"].score(b, False, True)
             nb = self.DRIFT_MAX_ROWS
         pvals = cpu_ref.pvals_from_stats(self.packed, g["np_hist"][0], g["np_ksd"][0], nb)
         resp = g["ext"].build_response_json(
@@ -309,7 +341,12 @@ class ScoringEngine:
         return {"response": resp, "latency_ms": latency_ms, "rows": len(codes)}
 
 
-def load_engine(model_directory: str, device: str = "auto", device_index: int = 0) -> ScoringEngine:
+def load_engine(
+    model_directory: str,
+    device: str = "auto",
+    device_index: int = 0,
+    drift_max_rows: int | None = None,
+) -> ScoringEngine:
     """Load a model into a ScoringEngine. Accepts a pyfunc model dir, a
     packed .npz, or a registry URI (``models:/<name>/<version|latest>`` —
     the reference's MLflow registry addressing, 02-register cell-15)."""
@@ -326,4 +363,9 @@ def load_engine(model_directory: str, device: str = "auto", device_index: int = 
         packed = PackedModel.load(model_directory)
     else:
         packed = packmod.pack_pyfunc_dir(model_directory)
-    return ScoringEngine(packed, device=device, device_index=device_index)
+    return ScoringEngine(
+        packed,
+        device=device,
+        device_index=device_index,
+        drift_max_rows=drift_max_rows,
+    )

@@ -261,6 +261,7 @@ class RawScoreServer:
                     break
                 clen = 0
                 keep_alive = True
+                admin_hdrs = {}
                 while True:
                     h = await reader.readline()
                     if h in (b"\r\n", b"\n", b""):
@@ -268,9 +269,20 @@ class RawScoreServer:
                     k, _, v = h.partition(b":")
                     lk = k.lower()
                     if lk == b"content-length":
-                        clen = int(v.strip())
+                        v = v.strip()
+                        # digits only: negative/malformed values must be a
+                        # 400, not an unhandled ValueError (and a negative
+                        # length would sail past the max_body_bytes check)
+                        if not v.isdigit():
+                            await self._respond(
+                                writer, 400, b'{"detail": "invalid Content-Length"}'
+                            )
+                            return
+                        clen = int(v)
                     elif lk == b"connection" and b"close" in v.lower():
                         keep_alive = False
+                    elif lk in (b"authorization", b"x-admin-token"):
+                        admin_hdrs[lk] = v.strip()
                     elif lk == b"transfer-encoding":
                         # chunked bodies: not needed by the contract clients
                         await self._respond(writer, 411, b'{"detail": "length required"}')
@@ -286,7 +298,23 @@ class RawScoreServer:
                 if method == b"POST" and ppath in (b"/score", b"/predict"):
                     status, payload = await self._score(body)
                 elif method == b"POST" and ppath == b"/admin/reload":
-                    status, payload = await self._reload(body)
+                    from .serve import admin_authorized
+
+                    peer = writer.get_extra_info("peername")
+                    auth = admin_hdrs.get(b"authorization")
+                    xtok = admin_hdrs.get(b"x-admin-token")
+                    if admin_authorized(
+                        self.cfg,
+                        peer[0] if peer else None,
+                        auth.decode("ascii", "replace") if auth else None,
+                        xtok.decode("ascii", "replace") if xtok else None,
+                    ):
+                        status, payload = await self._reload(body)
+                    else:
+                        status, payload = 403, (
+                            b'{"detail": "admin endpoint: loopback client'
+                            b' or admin token required"}'
+                        )
                 elif method == b"GET":
                     res = await self._get(path)
                     status, payload = res[0], res[1]
@@ -306,7 +334,7 @@ class RawScoreServer:
             except Exception:
                 pass
 
-    _REASONS = {200: b"OK", 400: b"Bad Request", 404: b"Not Found",
+    _REASONS = {200: b"OK", 400: b"Bad Request", 403: b"Forbidden", 404: b"Not Found",
                 405: b"Method Not Allowed", 411: b"Length Required",
                 413: b"Payload Too Large", 422: b"Unprocessable Entity",
                 500: b"Internal Server Error", 503: b"Service Unavailable"}

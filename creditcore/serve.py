@@ -40,15 +40,56 @@ state: dict = {}
 def _build_engines(cfg: ServeConfig) -> list[ScoringEngine]:
     device = cfg.resolve_device()
     if device == "cpu":
-        return [load_engine(cfg.model_directory, device="cpu")]
+        return [
+            load_engine(
+                cfg.model_directory, device="cpu", drift_max_rows=cfg.drift_max_batch
+            )
+        ]
     import torch
 
     n = min(cfg.n_gpus or torch.cuda.device_count(), torch.cuda.device_count())
-    first = load_engine(cfg.model_directory, device="cuda", device_index=0)
+    first = load_engine(
+        cfg.model_directory,
+        device="cuda",
+        device_index=0,
+        drift_max_rows=cfg.drift_max_batch,
+    )
     engines = [first]
     for i in range(1, n):
-        engines.append(ScoringEngine(first.packed, device="cuda", device_index=i))
+        engines.append(
+            ScoringEngine(
+                first.packed,
+                device="cuda",
+                device_index=i,
+                drift_max_rows=cfg.drift_max_batch,
+            )
+        )
     return engines
+
+
+# in-process test clients report these pseudo-hosts; real sockets give the
+# peer IP ("testclient" is FastAPI's TestClient, None = no transport info
+# from an in-process ASGI call)
+_LOOPBACK_HOSTS = {"127.0.0.1", "::1", "localhost", "testclient"}
+
+
+def admin_authorized(
+    cfg: ServeConfig,
+    client_host: str | None,
+    authorization: str | None = None,
+    x_admin_token: str | None = None,
+) -> bool:
+    """Gate for /admin/* (model hot-swap deserializes an arbitrary on-disk
+    path — it must not be open to the public listener). Token configured →
+    require it on every call; no token → loopback clients only."""
+    if cfg.admin_token:
+        presented = x_admin_token
+        if presented is None and authorization and authorization.lower().startswith("bearer "):
+            presented = authorization[7:]
+        import secrets
+
+        return presented is not None and secrets.compare_digest(presented, cfg.admin_token)
+    return client_host is None or client_host in _LOOPBACK_HOSTS
 
 
 class ReplicaPool:
@@ -387,6 +428,16 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         from .parallel import DriftSync
 
         cfg: ServeConfig = state["cfg"]
+        if not admin_authorized(
+            cfg,
+            request.client.host if request.client else None,
+            request.headers.get("authorization"),
+            request.headers.get("x-admin-token"),
+        ):
+            raise HTTPException(
+                status_code=403,
+                detail="admin endpoint: loopback client or admin token required",
+            )
         try:
             payload = json.loads(await request.body() or b"{}")
         except ValueError:

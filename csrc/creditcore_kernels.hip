@@ -1023,8 +1023,13 @@ struct ScoreSession {
     for (int j = 0; j + 1 < rs.size(0); ++j)
       max_ref_len = std::max(max_ref_len, (int64_t)(rs_acc[j + 1] - rs_acc[j]));
 
-    // opt in to >64 KiB dynamic LDS for the LDS-staged K-S path
+    // opt in to >64 KiB dynamic LDS for every launched K-S instantiation:
+    // record() launches ks_kernel_t<512> (b=16384 needs exactly 64 KiB of
+    // dynamic LDS, at/over the historical default cap), and the standalone
+    // drift_stats/ks_stats paths launch <256>.
     (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&ks_kernel_t<256>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)KS_LDS_BYTES);
+    (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&ks_kernel_t<512>),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)KS_LDS_BYTES);
 
     HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
@@ -1905,7 +1910,7 @@ py::tuple score_json_full(ScoreSession& s, py::bytes body,
                           const JsonEncoderState& st,
                           py::array_t<int32_t> ref_cat_counts,
                           py::array_t<int32_t> cat_offsets, int64_t n_ref,
-                          py::list feature_names) {
+                          py::list feature_names, int64_t drift_cap) {
   char* data;
   ssize_t blen;
   if (PyBytes_AsStringAndSize(body.ptr(), &data, &blen) != 0)
@@ -1932,12 +1937,22 @@ py::tuple score_json_full(ScoreSession& s, py::bytes body,
 
   std::memcpy(s.p_codes(0), codes.data(), b * N_CAT * sizeof(int16_t));
   std::memcpy(s.p_nums(0), nums.data(), b * N_NUM * sizeof(float));
-  const bool drift_now = (int64_t)b <= MAX_DRIFT_ROWS;
-  s.score((int64_t)b, drift_now, true, 0);
+  const int64_t cap = (drift_cap > 0) ? std::min<int64_t>(drift_cap, MAX_DRIFT_ROWS)
+                                      : MAX_DRIFT_ROWS;
   int64_t nb = (int64_t)b;
-  if (!drift_now) {  // oversized: capped drift sample in a second pass
-    s.score(MAX_DRIFT_ROWS, true, true, 0);
-    nb = MAX_DRIFT_ROWS;
+  if ((int64_t)b <= cap) {
+    s.score((int64_t)b, true, true, 0);
+  } else {
+    // Oversized batch: drift is a batch-population statistic, so run the
+    // capped-sample drift pass FIRST (only with_drift passes write the
+    // pinned drift region), then the full batch without drift — pin_outs
+    // then holds the b-packed layout the serializer below reads. Running
+    // the passes the other way round overwrote the b-packed outputs with a
+    // cap-packed layout and corrupted rows >= cap (round-1 advisor
+    // finding).
+    s.score(cap, true, true, 0);
+    s.score((int64_t)b, false, true, 0);
+    nb = cap;
   }
 
   double pv[N_CAT + N_NUM];
@@ -2066,7 +2081,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("synchronize", &ScoreSession::synchronize)
       .def("score_json_full", &score_json_full, py::arg("body"),
            py::arg("encoder"), py::arg("ref_cat_counts"),
-           py::arg("cat_offsets"), py::arg("n_ref"), py::arg("feature_names"))
+           py::arg("cat_offsets"), py::arg("n_ref"), py::arg("feature_names"),
+           py::arg("drift_cap") = -1)
       .def_readonly("capacity", &ScoreSession::capacity)
       .def_property_readonly("pin_codes", [](ScoreSession& s) { return s.pin_codes; })
       .def_property_readonly("pin_nums", [](ScoreSession& s) { return s.pin_nums; })

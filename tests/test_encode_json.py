@@ -231,3 +231,32 @@ def test_c_serializer_double_roundtrip(ext):
     assert doc["outliers"] == [float(i % 2) for i in range(b)]
     for f, p in zip(FEATURES, pvals):
         assert doc["feature_drift_batch"][f] == float(np.float32(1.0) - np.float32(p))
+
+
+def test_cpu_engine_json_body_fills_schema_defaults(packed, monkeypatch):
+    """device=cpu encode_json_body (extension absent) must fill absent
+    fields with the schema defaults (pydantic default semantics, reference
+    app/model.py:8-34) — not MISSING_CATEGORY (round-1 advisor low
+    finding)."""
+    import json
+
+    from creditcore.engine import ScoringEngine
+    from creditcore.ops import gpu
+    from creditcore.pack import encode_batch
+    from creditcore.schema import LoanApplicant
+
+    monkeypatch.setattr(gpu, "available", lambda: False)
+    eng = ScoringEngine(packed, device="cpu")
+    partial = [{"credit_limit": 123.0}, {"sex": "female", "age": 40.0}]
+    codes, nums = eng.encode_json_body(json.dumps(partial).encode())
+    full = [{**LoanApplicant().__dict__, **r} for r in partial]
+    codes_ref, nums_ref = encode_batch(full, packed.vocabs)
+    np.testing.assert_array_equal(codes, codes_ref)
+    np.testing.assert_array_equal(nums, nums_ref)
+    # malformed shapes raise ValueError (callers map it to 4xx)
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        eng.encode_json_body(b'{"not": "a list"}')
+    with _pytest.raises(ValueError):
+        eng.encode_json_body(b'[1, 2]')

@@ -57,7 +57,7 @@ def test_drift_parity(gpu_engine, packed, encoded):
     np.testing.assert_allclose(out["p_vals"], pvals, atol=1e-6)
 
 
-@pytest.mark.parametrize("b", [1, 2, 63, 64, 65, 1024, 4096])
+@pytest.mark.parametrize("b", [1, 2, 63, 64, 65, 1024, 4096, 8192, 16384])
 def test_batch_size_sweep(gpu_engine, packed, b):
     rng = np.random.default_rng(b)
     codes = np.stack(
@@ -152,6 +152,84 @@ def test_serving_stack_on_gpu(model_dir):
         assert client.get("/healthz").json()["status"] == "ok"
         drift = client.get("/drift").json()
         assert drift["rows"] >= 301
+
+
+def _random_encoded(packed, b, seed=0):
+    rng = np.random.default_rng(seed)
+    codes = np.stack(
+        [rng.integers(-1, len(v), size=b) for v in packed.vocabs], axis=1
+    ).astype(np.int16)
+    nums = rng.normal(5000.0, 3000.0, size=(b, 14)).astype(np.float32)
+    return codes, nums
+
+
+def test_oversized_batch_bytes_path(gpu_engine, packed):
+    """b > DRIFT_MAX_ROWS through score_encoded_bytes: predictions for every
+    row (including rows >= 16384) must match the CPU reference — guards the
+    round-1 advisor finding where the capped drift second pass overwrote the
+    b-packed pinned outputs before serialization."""
+    import json
+
+    b = 20000
+    codes, nums = _random_encoded(packed, b, seed=7)
+    out = gpu_engine.score_encoded_bytes(codes, nums)
+    assert out["rows"] == b
+    resp = json.loads(out["response_bytes"])
+    assert len(resp["predictions"]) == b and len(resp["outliers"]) == b
+    ref = cpu_ref.score_batch_cpu(packed, codes, nums)
+    np.testing.assert_allclose(resp["predictions"], ref["predictions"], atol=1e-9)
+    np.testing.assert_array_equal(resp["outliers"], ref["outliers"])
+    # drift block reflects the capped 16384-row sample
+    chists, ks_d = cpu_ref.drift_stats_cpu(
+        packed,
+        codes[: gpu_engine.DRIFT_MAX_ROWS],
+        cpu_ref.impute_nums(packed, nums[: gpu_engine.DRIFT_MAX_ROWS]),
+    )
+    pvals = cpu_ref.pvals_from_stats(packed, chists, ks_d, gpu_engine.DRIFT_MAX_ROWS)
+    from creditcore.schema import FEATURES
+
+    got = np.array([resp["feature_drift_batch"][f] for f in FEATURES])
+    np.testing.assert_allclose(got, 1.0 - np.asarray(pvals, dtype=np.float32),
+                               atol=1e-6)
+
+
+def test_oversized_batch_json_full_path(gpu_engine, packed):
+    """Same guard for the one-call C++ path (score_json_full at
+    csrc/creditcore_kernels.hip score_json_full): grow capacity first so the
+    C++ oversized branch (not the Python fallback) executes."""
+    import json
+
+    from creditcore.data import make_request_batch
+
+    b = 20000
+    # grow the session past b so score_json_full takes the C++ branch
+    gpu_engine._ensure_capacity(b)
+    body = json.dumps(make_request_batch(b, seed=3)).encode()
+    out = gpu_engine.score_json_full(body)
+    assert out["rows"] == b
+    resp = json.loads(out["response_bytes"])
+    assert len(resp["predictions"]) == b
+    codes, nums = gpu_engine.encode_json_body(body)
+    ref = cpu_ref.score_batch_cpu(packed, codes, nums)
+    np.testing.assert_allclose(resp["predictions"], ref["predictions"], atol=1e-9)
+    np.testing.assert_array_equal(resp["outliers"], ref["outliers"])
+
+
+def test_drift_cap_config_respected(packed):
+    """A configured drift_max_batch below the hardware ceiling caps the
+    drift sample while every row is still scored (config wiring —
+    round-1 advisor low finding)."""
+    import json
+
+    eng = ScoringEngine(packed, device="cuda", device_index=0, drift_max_rows=4096)
+    assert eng.DRIFT_MAX_ROWS == 4096
+    b = 6000
+    codes, nums = _random_encoded(packed, b, seed=9)
+    out = eng.score_encoded_bytes(codes, nums)
+    resp = json.loads(out["response_bytes"])
+    assert len(resp["predictions"]) == b
+    ref = cpu_ref.score_batch_cpu(packed, codes, nums)
+    np.testing.assert_allclose(resp["predictions"], ref["predictions"], atol=1e-9)
 
 
 def test_native_code_is_loaded(gpu_engine):

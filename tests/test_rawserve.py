@@ -118,6 +118,90 @@ def test_admin_reload_on_raw_frontend(raw_url, model_dir):
     assert still.status_code == 200
 
 
+def test_invalid_content_length_gets_400(raw_url):
+    """Malformed / negative Content-Length must be a clean 400, not an
+    unhandled ValueError in the connection task (round-1 advisor low
+    finding; a negative value would also pass the max_body_bytes check)."""
+    import socket
+
+    host, port = raw_url.rsplit("/", 1)[-1].split(":")
+    for bad in (b"-5", b"abc", b"1e3", b" "):
+        with socket.create_connection((host, int(port)), timeout=30) as s:
+            s.sendall(
+                b"POST /score HTTP/1.1\r\nHost: x\r\nContent-Length: "
+                + bad
+                + b"\r\n\r\n"
+            )
+            resp = s.recv(4096)
+        assert resp.startswith(b"HTTP/1.1 400"), (bad, resp[:80])
+
+
+def test_admin_reload_requires_token_when_configured(model_dir):
+    """With admin_token set, /admin/reload rejects calls without the token
+    even from loopback, and accepts X-Admin-Token (advisor medium
+    finding: unauthenticated model hot-swap on the public listener)."""
+    import socket
+    import threading
+
+    import httpx
+
+    from creditcore.config import ServeConfig
+    from creditcore.rawserve import RawScoreServer
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    cfg.host = "127.0.0.1"
+    cfg.port = port
+    cfg.admin_token = "sekrit"
+
+    loop = asyncio.new_event_loop()
+    server = RawScoreServer(cfg)
+    started = threading.Event()
+
+    def run():
+        asyncio.set_event_loop(loop)
+
+        async def go():
+            await server.start()
+            started.set()
+            await server._server.serve_forever()
+
+        try:
+            loop.run_until_complete(go())
+        except (asyncio.CancelledError, RuntimeError):
+            pass
+
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    assert started.wait(timeout=120)
+    try:
+        url = f"http://127.0.0.1:{port}"
+        body = {"model_uri": model_dir}
+        assert httpx.post(f"{url}/admin/reload", json=body, timeout=60).status_code == 403
+        r = httpx.post(
+            f"{url}/admin/reload",
+            json=body,
+            headers={"X-Admin-Token": "sekrit"},
+            timeout=120,
+        )
+        assert r.status_code == 200 and r.json()["status"] == "reloaded"
+        r = httpx.post(
+            f"{url}/admin/reload",
+            json=body,
+            headers={"Authorization": "Bearer wrong"},
+            timeout=60,
+        )
+        assert r.status_code == 403
+    finally:
+        loop.call_soon_threadsafe(loop.stop)
+        t.join(timeout=30)
+
+
 def test_oversized_body_gets_413(raw_url):
     """A hostile Content-Length must be refused before the read, not
     buffered into memory."""
