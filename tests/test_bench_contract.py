@@ -84,3 +84,38 @@ def test_bench_torchrun_ws2_gloo(tmp_path):
     assert d["config"]["global_batch"] == 2 * 64
     # exactly one result line (rank 0 only)
     assert sum(1 for l in r.stdout.splitlines() if l.startswith("{")) == 1
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(600)
+def test_bench_gpus_n_self_spawns_real_ranks(tmp_path):
+    """`bench.py --gpus 2` WITHOUT torchrun must launch 2 real ranks and
+    report their measured aggregate — never multiply a single-rank rate by
+    N (round-1 verdict weak-spot #1)."""
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--device", "cpu",
+         "--gpus", "2", "--steps", "2", "--warmup", "1", "--rows", "64",
+         "--model-trees", "12", "--model-depth", "4"],
+        capture_output=True, text=True, timeout=540, env=_env(tmp_path), cwd=REPO,
+    )
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    assert "self-spawning 2 ranks" in r.stderr
+    d = _last_json_line(r.stdout)
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert sum(1 for l in r.stdout.splitlines() if l.startswith("{")) == 1
+
+
+@pytest.mark.timeout(300)
+def test_bench_refuses_to_extrapolate_gpus(tmp_path):
+    """--gpus N with cuda requested but N devices unavailable must refuse
+    (exit 2), not print an extrapolated whole-node number."""
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--device", "cuda",
+         "--gpus", "8", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=240, env=_env(tmp_path), cwd=REPO,
+    )
+    # no GPU in CI: device_count()==0 < 8 -> refusal path
+    assert r.returncode == 2, (r.returncode, r.stdout[-500:], r.stderr[-500:])
+    assert "refusing to extrapolate" in r.stderr
+    assert not [l for l in r.stdout.splitlines() if l.startswith("{")]
