@@ -48,6 +48,12 @@ class ServeConfig:
 
     # drift
     drift_sync_period: int = field(default_factory=lambda: _env("drift_sync_period", 64, int))
+    # replica health: period of the probation loop that re-probes dead
+    # replicas and re-admits the ones that answer (SURVEY §5.3 recovery)
+    replica_probe_period_s: float = field(
+        default_factory=lambda: _env("replica_probe_period_s", 5.0, float)
+    )
+
     # drift-sample cap per request batch (rows beyond it are still scored;
     # drift statistics use the first drift_max_batch rows). Hardware ceiling
     # is 16384 (the K-S kernel's LDS sort capacity); values above clamp.

@@ -90,6 +90,19 @@ class RawScoreServer:
         for b in self.batchers:
             await b.start()
         self.pool = ReplicaPool(len(self.engines))
+
+        # probation loop: re-probe dead replicas, re-admit on success
+        async def _revival_loop():
+            from .serve import probe_revive
+
+            while True:
+                await asyncio.sleep(max(cfg.replica_probe_period_s, 0.25))
+                try:
+                    await probe_revive(self.pool, self.engines)
+                except Exception:
+                    pass
+
+        self._revival_task = asyncio.ensure_future(_revival_loop())
         self._server = await asyncio.start_server(
             self._handle, cfg.host, cfg.port, backlog=512,
             reuse_port=(cfg.workers > 1)
@@ -98,6 +111,8 @@ class RawScoreServer:
         return self._server
 
     async def close(self):
+        if getattr(self, "_revival_task", None) is not None:
+            self._revival_task.cancel()
         self._server.close()
         await self._server.wait_closed()
         for b in self.batchers:
@@ -129,12 +144,7 @@ class RawScoreServer:
             reqlog.log_inference_data(
                 cfg.service_name, request_id, body.decode("utf-8", "replace")
             )
-        try:
-            idx = self.pool.pick()
-        except RuntimeError:
-            self.metrics.observe_error()
-            return 503, b'{"detail": "no healthy replicas"}'
-        engine = self.engines[idx]
+        engine = self.engines[0]
         loop = asyncio.get_running_loop()
         try:
             codes, nums = await loop.run_in_executor(
@@ -154,14 +164,28 @@ class RawScoreServer:
         if len(codes) == 0:
             return 400, b'{"detail": "empty request batch"}'
 
+        # failover across replicas: one failing GPU must not fail requests
+        # while healthy replicas remain (see serve._predict_impl)
         t0 = time.perf_counter()
-        try:
-            out = await self.batchers[idx].submit(codes, nums)
-            self.pool.report_ok(idx)
-        except Exception as e:
-            self.metrics.observe_error()
-            self.pool.report_fail(idx)
-            return 500, json.dumps({"detail": f"scoring failed: {e}"}).encode()
+        out = None
+        last_exc = None
+        for _ in range(self.pool.attempt_budget):
+            try:
+                idx = self.pool.pick()
+            except RuntimeError:
+                self.metrics.observe_error()
+                return 503, b'{"detail": "no healthy replicas"}'
+            engine = self.engines[idx]
+            try:
+                out = await self.batchers[idx].submit(codes, nums)
+                self.pool.report_ok(idx)
+                break
+            except Exception as e:
+                self.metrics.observe_error()
+                self.pool.report_fail(idx)
+                last_exc = e
+        if out is None:
+            return 500, json.dumps({"detail": f"scoring failed: {last_exc}"}).encode()
         latency_ms = (time.perf_counter() - t0) * 1e3
         self.metrics.observe_request(len(codes), latency_ms)
 

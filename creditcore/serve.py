@@ -38,16 +38,24 @@ state: dict = {}
 
 
 def _build_engines(cfg: ServeConfig) -> list[ScoringEngine]:
+    """One ScoringEngine replica per serving slot. n_gpus picks the replica
+    count (0 = one per visible GPU); asking for more replicas than devices
+    oversubscribes round-robin (multiple independent sessions per GPU —
+    used by failover tests and latency isolation)."""
     device = cfg.resolve_device()
     if device == "cpu":
-        return [
-            load_engine(
-                cfg.model_directory, device="cpu", drift_max_rows=cfg.drift_max_batch
-            )
+        n = max(cfg.n_gpus, 1)
+        first = load_engine(
+            cfg.model_directory, device="cpu", drift_max_rows=cfg.drift_max_batch
+        )
+        return [first] + [
+            ScoringEngine(first.packed, device="cpu", drift_max_rows=cfg.drift_max_batch)
+            for _ in range(n - 1)
         ]
     import torch
 
-    n = min(cfg.n_gpus or torch.cuda.device_count(), torch.cuda.device_count())
+    n_dev = max(torch.cuda.device_count(), 1)
+    n = cfg.n_gpus or n_dev
     first = load_engine(
         cfg.model_directory,
         device="cuda",
@@ -60,7 +68,7 @@ def _build_engines(cfg: ServeConfig) -> list[ScoringEngine]:
             ScoringEngine(
                 first.packed,
                 device="cuda",
-                device_index=i,
+                device_index=i % n_dev,
                 drift_max_rows=cfg.drift_max_batch,
             )
         )
@@ -119,6 +127,50 @@ class ReplicaPool:
         self.fails[i] += 1
         if self.fails[i] >= self.max_failures:
             self.alive[i] = False
+
+    def revive(self, i: int) -> None:
+        """Re-admit a dead replica after a successful probation probe (the
+        recovery half of SURVEY.md §5.3 — detection/redistribution alone
+        would shrink the pool monotonically until a full reload)."""
+        self.fails[i] = 0
+        self.alive[i] = True
+
+    def dead_indices(self) -> list[int]:
+        return [i for i, a in enumerate(self.alive) if not a]
+
+    @property
+    def attempt_budget(self) -> int:
+        """Upper bound on per-request failover attempts that guarantees a
+        healthy replica is reached if one exists (round-robin pick can hit
+        a failing replica at most max_failures times before it drops)."""
+        return len(self.alive) * self.max_failures + 1
+
+
+async def probe_revive(pool: ReplicaPool, engines, logger=None) -> list[int]:
+    """Probe every dead replica with the schema-default record and re-admit
+    the ones that answer. Returns the revived indices."""
+    import asyncio
+
+    dead = pool.dead_indices()
+    if not dead:
+        return []
+    codes, nums = encode_batch(
+        [LoanApplicant().__dict__], engines[0].packed.vocabs
+    )
+    loop = asyncio.get_running_loop()
+    revived = []
+    for i in dead:
+        try:
+            await loop.run_in_executor(
+                None, lambda e=engines[i]: e.score_arrays(codes, nums, False)
+            )
+            pool.revive(i)
+            revived.append(i)
+            if logger is not None:
+                logger.info("replica %d revived after probation probe", i)
+        except Exception:
+            pass  # still dead; next period re-probes
+    return revived
 
 
 def _make_batchers(cfg: ServeConfig, engines, drift_sync) -> list[MicroBatcher]:
@@ -194,7 +246,29 @@ async def lifespan(app: FastAPI):
     state["drift_sync"] = drift_sync
     state["metrics"] = Metrics()
     state["cfg"] = cfg
+
+    # probation loop: periodically re-probe dead replicas and re-admit the
+    # ones that recover (reads state each tick, so /admin/reload swaps are
+    # picked up)
+    import asyncio
+    import logging as _logging
+
+    async def _revival_loop():
+        log = _logging.getLogger("creditcore.replicas")
+        while True:
+            await asyncio.sleep(max(cfg.replica_probe_period_s, 0.25))
+            pool = state.get("pool")
+            live_engines = state.get("engines")
+            if pool is None or not live_engines:
+                continue
+            try:
+                await probe_revive(pool, live_engines, log)
+            except Exception:
+                pass
+
+    revival_task = asyncio.create_task(_revival_loop())
     yield
+    revival_task.cancel()
     # use the *current* objects — /admin/reload may have swapped them
     for b in state.get("batchers", batchers):
         await b.close()
@@ -242,12 +316,7 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         batchers = state["batchers"]
         pool: ReplicaPool = state["pool"]
         try:
-            idx = pool.pick()
-        except RuntimeError:
-            metrics.observe_error()
-            raise HTTPException(status_code=503, detail="no healthy replicas")
-        try:
-            codes, nums = _encode_body(body, engines[idx])
+            codes, nums = _encode_body(body, engines[0])
         except HTTPException:
             raise
         except (ValueError, TypeError) as e:
@@ -255,14 +324,28 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         if len(codes) == 0:
             raise HTTPException(status_code=400, detail="empty request batch")
 
+        # failover: a replica failing mid-flight must not fail the request
+        # while healthy replicas exist (the K8s-Service-retries analog);
+        # the attempt budget guarantees a healthy replica is reached if any
         t0 = time.perf_counter()
-        try:
-            out = await batchers[idx].submit(codes, nums)
-            pool.report_ok(idx)
-        except Exception as e:
-            metrics.observe_error()
-            pool.report_fail(idx)
-            raise HTTPException(status_code=500, detail=f"scoring failed: {e}")
+        out = None
+        last_exc: Exception | None = None
+        for _ in range(pool.attempt_budget):
+            try:
+                idx = pool.pick()
+            except RuntimeError:
+                metrics.observe_error()
+                raise HTTPException(status_code=503, detail="no healthy replicas")
+            try:
+                out = await batchers[idx].submit(codes, nums)
+                pool.report_ok(idx)
+                break
+            except Exception as e:
+                metrics.observe_error()
+                pool.report_fail(idx)
+                last_exc = e
+        if out is None:
+            raise HTTPException(status_code=500, detail=f"scoring failed: {last_exc}")
         latency_ms = (time.perf_counter() - t0) * 1e3
 
         if "response_bytes" in out:  # solo-flush wire-out fast path

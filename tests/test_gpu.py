@@ -232,6 +232,55 @@ def test_drift_cap_config_respected(packed):
     np.testing.assert_allclose(resp["predictions"], ref["predictions"], atol=1e-9)
 
 
+def test_gpu_fault_injection_failover_and_recovery(model_dir, monkeypatch):
+    """Poison one GPU replica mid-traffic (faults surface exactly like HIP
+    errors do — RuntimeError out of the scoring call): requests keep
+    answering 200 on the survivor, the poisoned replica leaves rotation,
+    and the probation loop re-admits it once it answers again
+    (SURVEY §5.3 detection + recovery on hardware)."""
+    import time as _time
+
+    from fastapi.testclient import TestClient
+
+    from creditcore.config import ServeConfig
+    from creditcore.schema import SAMPLE_REQUEST
+    from creditcore.serve import create_app, state
+
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.device = "cuda"
+    cfg.n_gpus = 2  # two independent sessions (oversubscribed on 1 GPU)
+    cfg.replica_probe_period_s = 0.3
+    with TestClient(create_app(cfg)) as c:
+        assert len(state["engines"]) == 2
+        e0 = state["engines"][0]
+        b0 = state["batchers"][0]
+        orig_arrays = e0.score_arrays
+        orig_barrays = b0.score_arrays
+        orig_bsingle = b0.score_single
+
+        def boom(*a, **k):
+            raise RuntimeError("hipErrorInjected: device fault (test)")
+
+        monkeypatch.setattr(e0, "score_arrays", boom)
+        monkeypatch.setattr(b0, "score_arrays", boom)
+        if b0.score_single is not None:
+            monkeypatch.setattr(b0, "score_single", boom)
+        for _ in range(8):
+            assert c.post("/predict", json=SAMPLE_REQUEST).status_code == 200
+        assert state["pool"].alive == [False, True]
+
+        monkeypatch.setattr(e0, "score_arrays", orig_arrays)
+        monkeypatch.setattr(b0, "score_arrays", orig_barrays)
+        monkeypatch.setattr(b0, "score_single", orig_bsingle)
+        deadline = _time.time() + 20
+        while _time.time() < deadline and not all(state["pool"].alive):
+            _time.sleep(0.1)
+        assert all(state["pool"].alive), "replica 0 never revived"
+        for _ in range(4):
+            assert c.post("/predict", json=SAMPLE_REQUEST).status_code == 200
+
+
 def test_native_code_is_loaded(gpu_engine):
     """The loaded extension must be the in-tree .so (native-code check)."""
     import creditcore._ccore as ccore
