@@ -273,6 +273,42 @@ class RawScoreServer:
         return 404, b'{"detail": "not found"}'
 
     # ------------------------------------------------------------ HTTP/1.1
+    async def _read_chunked(self, reader):
+        """Decode a Transfer-Encoding: chunked body (RFC 9112 §7.1) so any
+        standard HTTP client works, matching the reference's
+        accept-anything serving contract (reference app/main.py:42).
+        Returns (body, None) or (None, (status, payload))."""
+        parts = []
+        total = 0
+        cap = self.cfg.max_body_bytes
+        while True:
+            size_line = await reader.readline()
+            if not size_line:
+                return None, (400, b'{"detail": "truncated chunked body"}')
+            # chunk extensions after ';' are ignored per spec
+            size_tok = size_line.split(b";", 1)[0].strip()
+            try:
+                size = int(size_tok, 16)
+            except ValueError:
+                return None, (400, b'{"detail": "invalid chunk size"}')
+            if size < 0:
+                return None, (400, b'{"detail": "invalid chunk size"}')
+            if size == 0:
+                break
+            total += size
+            if total > cap:
+                return None, (413, b'{"detail": "body too large"}')
+            parts.append(await reader.readexactly(size))
+            crlf = await reader.readexactly(2)
+            if crlf != b"\r\n":
+                return None, (400, b'{"detail": "malformed chunk"}')
+        # trailer section: read until the blank line
+        while True:
+            t = await reader.readline()
+            if t in (b"\r\n", b"\n", b""):
+                break
+        return b"".join(parts), None
+
     async def _handle(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
         try:
             while True:
@@ -285,6 +321,7 @@ class RawScoreServer:
                     break
                 clen = 0
                 keep_alive = True
+                chunked = False
                 admin_hdrs = {}
                 while True:
                     h = await reader.readline()
@@ -308,15 +345,27 @@ class RawScoreServer:
                     elif lk in (b"authorization", b"x-admin-token"):
                         admin_hdrs[lk] = v.strip()
                     elif lk == b"transfer-encoding":
-                        # chunked bodies: not needed by the contract clients
-                        await self._respond(writer, 411, b'{"detail": "length required"}')
-                        return
+                        te = v.strip().lower()
+                        if te == b"chunked":
+                            chunked = True
+                        else:
+                            # compressed transfer codings are out of contract
+                            await self._respond(
+                                writer, 501, b'{"detail": "unsupported transfer-encoding"}'
+                            )
+                            return
                 if clen > self.cfg.max_body_bytes:
                     # refuse before reading: an unbounded Content-Length must
                     # not drive readexactly into allocating it
                     await self._respond(writer, 413, b'{"detail": "body too large"}')
                     return
-                body = await reader.readexactly(clen) if clen else b""
+                if chunked:
+                    body, err = await self._read_chunked(reader)
+                    if err is not None:
+                        await self._respond(writer, err[0], err[1])
+                        return
+                else:
+                    body = await reader.readexactly(clen) if clen else b""
                 ctype = b"application/json"
                 ppath = path.partition(b"?")[0]
                 if method == b"POST" and ppath in (b"/score", b"/predict"):
@@ -361,6 +410,7 @@ class RawScoreServer:
     _REASONS = {200: b"OK", 400: b"Bad Request", 403: b"Forbidden", 404: b"Not Found",
                 405: b"Method Not Allowed", 411: b"Length Required",
                 413: b"Payload Too Large", 422: b"Unprocessable Entity",
+                501: b"Not Implemented",
                 500: b"Internal Server Error", 503: b"Service Unavailable"}
 
     async def _respond(self, writer, status: int, payload: bytes,

@@ -217,21 +217,6 @@ def test_oversized_body_gets_413(raw_url):
     assert data.startswith(b"HTTP/1.1 413")
 
 
-def test_chunked_body_gets_411(raw_url):
-    """Transfer-Encoding: chunked is outside the contract clients' subset —
-    the frontend must refuse it deterministically, not misparse it."""
-    import socket
-
-    host, port = raw_url[len("http://"):].rsplit(":", 1)
-    with socket.create_connection((host, int(port)), timeout=15) as s:
-        s.sendall(
-            b"POST /score HTTP/1.1\r\nHost: x\r\n"
-            b"Transfer-Encoding: chunked\r\n\r\n"
-        )
-        data = s.recv(4096)
-    assert data.startswith(b"HTTP/1.1 411")
-
-
 @pytest.mark.slow
 @pytest.mark.timeout(180)
 def test_multi_worker_reuseport_lifecycle(model_dir, tmp_path):
@@ -317,3 +302,79 @@ def test_raw_replica_failure_gives_503(raw_url, model_dir):
     # hammering a healthy server (alive flags visible in /healthz)
     h = httpx.get(f"{raw_url}/healthz").json()
     assert h["alive"] == [True]
+
+
+def test_chunked_transfer_encoding(raw_url):
+    """A standard chunked client round-trips (round-1: chunked got 411;
+    the reference contract accepts any standard HTTP client)."""
+    import httpx
+
+    from creditcore.data import make_request_batch
+    from creditcore.schema import ModelOutput
+
+    body = json.dumps(make_request_batch(16, seed=8)).encode()
+
+    def gen():  # httpx sends a generator body as Transfer-Encoding: chunked
+        for i in range(0, len(body), 37):
+            yield body[i : i + 37]
+
+    r = httpx.post(
+        f"{raw_url}/score",
+        content=gen(),
+        headers={"Content-Type": "application/json"},
+        timeout=60.0,
+    )
+    assert r.status_code == 200, r.text
+    out = ModelOutput.model_validate(r.json())
+    assert len(out.predictions) == 16
+
+
+def test_chunked_extensions_trailers_and_limits(raw_url):
+    """Spec edges: chunk extensions and trailers are tolerated; an
+    oversized chunked body is refused with 413; garbage chunk sizes are
+    400."""
+    import socket
+
+    from creditcore.schema import SAMPLE_REQUEST
+
+    host, port = raw_url.rsplit("/", 1)[-1].split(":")
+    body = json.dumps(SAMPLE_REQUEST).encode()
+    half = len(body) // 2
+
+    def send(raw: bytes) -> bytes:
+        with socket.create_connection((host, int(port)), timeout=30) as s:
+            s.sendall(raw)
+            s.settimeout(30)
+            out = b""
+            while b"\r\n\r\n" not in out or len(out) < 16:
+                chunk = s.recv(65536)
+                if not chunk:
+                    break
+                out += chunk
+            return out
+
+    # extensions after ';' + a trailer header
+    msg = (
+        b"POST /score HTTP/1.1\r\nHost: x\r\nTransfer-Encoding: chunked\r\n\r\n"
+        + format(half, "x").encode() + b";ext=1\r\n" + body[:half] + b"\r\n"
+        + format(len(body) - half, "x").encode() + b"\r\n" + body[half:] + b"\r\n"
+        + b"0\r\nX-Trailer: t\r\n\r\n"
+    )
+    assert send(msg).startswith(b"HTTP/1.1 200")
+
+    bad = (
+        b"POST /score HTTP/1.1\r\nHost: x\r\nTransfer-Encoding: chunked\r\n\r\n"
+        b"zz\r\n"
+    )
+    assert send(bad).startswith(b"HTTP/1.1 400")
+
+    huge = (
+        b"POST /score HTTP/1.1\r\nHost: x\r\nTransfer-Encoding: chunked\r\n\r\n"
+        b"ffffffffff\r\n"
+    )
+    assert send(huge).startswith(b"HTTP/1.1 413")
+
+    gz = (
+        b"POST /score HTTP/1.1\r\nHost: x\r\nTransfer-Encoding: gzip\r\n\r\n"
+    )
+    assert send(gz).startswith(b"HTTP/1.1 501")
