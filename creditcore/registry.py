@@ -185,6 +185,33 @@ def save_pyfunc_model(
     with open(os.path.join(model_dir, "conda.yaml"), "w") as f:
         yaml.safe_dump(conda_env, f, sort_keys=False)
 
+    # the rest of mlflow's standard file set (the MLmodel env block above
+    # references python_env.yaml; real mlflow writes all three)
+    import sys
+
+    pyver = ".".join(map(str, sys.version_info[:3]))
+    reqs = [
+        "numpy",
+        "pandas",
+        "scikit-learn",
+        "scipy",
+        "joblib",
+        f"cloudpickle=={cloudpickle.__version__}",
+        "creditcore",
+    ]
+    with open(os.path.join(model_dir, "python_env.yaml"), "w") as f:
+        yaml.safe_dump(
+            {
+                "python": pyver,
+                "build_dependencies": ["pip", "setuptools", "wheel"],
+                "dependencies": ["-r requirements.txt"],
+            },
+            f,
+            sort_keys=False,
+        )
+    with open(os.path.join(model_dir, "requirements.txt"), "w") as f:
+        f.write("\n".join(reqs) + "\n")
+
     return model_dir
 
 
