@@ -86,20 +86,6 @@ def _evaluate(
     return EvalRun(uuid.uuid4().hex, dict(params), metrics, estimator)
 
 
-def _sample_params(rng: np.random.Generator, incumbent: dict | None) -> dict:
-    """Search-space draw (space: reference cell-8). With an incumbent, draw a
-    local perturbation half the time (TPE-style exploitation)."""
-    if incumbent is not None and rng.uniform() < 0.5:
-        n_est = int(np.clip(incumbent["n_estimators"] + rng.integers(-150, 151), 100, 999))
-        depth = int(np.clip(incumbent["max_depth"] + rng.integers(-4, 5), 1, 24))
-        crit = incumbent["criterion"] if rng.uniform() < 0.8 else rng.choice(["gini", "entropy"])
-    else:
-        n_est = int(rng.integers(100, 1000))
-        depth = int(rng.integers(1, 25))
-        crit = str(rng.choice(["gini", "entropy"]))
-    return {"n_estimators": n_est, "max_depth": depth, "criterion": str(crit)}
-
-
 def train_model(
     df: pd.DataFrame | None = None,
     max_evals: int = 10,
@@ -109,17 +95,22 @@ def train_model(
     n_rows: int = 20_000,
     algorithm: str = "rf",
 ) -> EvalRun:
-    """Hyperparameter search; returns the best run by roc_auc
-    (reference cell-8/10)."""
+    """Hyperparameter search with a real TPE sampler (reference cell-8:
+    hyperopt fmin(tpe.suggest, max_evals=10); hyperopt itself is
+    unavailable offline — creditcore.models.tpe implements the algorithm).
+    Returns the best run by roc_auc (cell-10)."""
+    from .models.tpe import TPESampler, reference_space
+
     if df is None:
         df = make_uci_shaped_frame(n_rows=n_rows, seed=seed)
-    rng = np.random.default_rng(seed)
+    sampler = TPESampler(reference_space(), seed=seed, n_startup=n_startup)
     best: EvalRun | None = None
     t0 = time.time()
     for i in range(max_evals):
-        incumbent = best.params if (best is not None and i >= n_startup) else None
-        params = _sample_params(rng, incumbent)
+        params = sampler.suggest()
         run = _evaluate(params, df, seed=seed, algorithm=algorithm)
+        # hyperopt minimizes; the notebook returns -roc_auc as the loss
+        sampler.observe(params, -run.metrics["validation_roc_auc_score"])
         if runs_dir:
             rd = os.path.join(runs_dir, run.run_id)
             os.makedirs(rd, exist_ok=True)
