@@ -28,7 +28,7 @@ import numpy as np
 @dataclass
 class _Pending:
     codes: np.ndarray
-    nums: np.ndarray
+    nums: np.ndarray | None  # None for single-array payloads (dense family)
     future: asyncio.Future = field(repr=False)
 
 
@@ -134,7 +134,11 @@ class MicroBatcher:
                 p.future.set_result(out)
             return
         codes = np.concatenate([p.codes for p in batch], axis=0)
-        nums = np.concatenate([p.nums for p in batch], axis=0)
+        nums = (
+            np.concatenate([p.nums for p in batch], axis=0)
+            if batch[0].nums is not None
+            else None
+        )
         loop = asyncio.get_running_loop()
         try:
             out = await loop.run_in_executor(None, self.score_arrays, codes, nums)

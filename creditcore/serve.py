@@ -232,14 +232,48 @@ async def lifespan(app: FastAPI):
     batchers = _make_batchers(cfg, engines, drift_sync)
     for b in batchers:
         await b.start()
-    dense_engine = None
+    # Dense wide-tabular family (BASELINE config 5) as a first-class served
+    # path: one DenseEngine replica per GPU (each holding the full sorted
+    # drift reference in its own HBM — the 8×40 GB data-parallel sizing
+    # story), micro-batched and pool-dispatched like the credit path.
+    dense_engines: list = []
+    dense_batchers: list = []
     if cfg.dense_model_dir:
         from .dense import DenseEngine, DenseModel
 
-        dense_engine = DenseEngine(
-            DenseModel.load(cfg.dense_model_dir), device=cfg.resolve_device()
-        )
-    state["dense_engine"] = dense_engine
+        dm = DenseModel.load(cfg.dense_model_dir)
+        device = cfg.resolve_device()
+        if device == "cpu":
+            dense_engines = [DenseEngine(dm, device="cpu")]
+        else:
+            import torch
+
+            n_dev = max(torch.cuda.device_count(), 1)
+            n = cfg.n_gpus or n_dev
+            dense_engines = [
+                DenseEngine(dm, device="cuda", device_index=i % n_dev)
+                for i in range(n)
+            ]
+
+        def _dense_scorer(e):
+            def run(x, _nums):
+                return e.score_arrays(x)
+
+            return run
+
+        dense_batchers = [
+            MicroBatcher(
+                _dense_scorer(e),
+                max_rows=cfg.max_batch_rows,
+                max_wait_us=cfg.batch_wait_us,
+            )
+            for e in dense_engines
+        ]
+        for b in dense_batchers:
+            await b.start()
+    state["dense_engines"] = dense_engines
+    state["dense_batchers"] = dense_batchers
+    state["dense_pool"] = ReplicaPool(len(dense_engines)) if dense_engines else None
     state["engines"] = engines
     state["batchers"] = batchers
     state["pool"] = ReplicaPool(len(engines))
@@ -271,6 +305,8 @@ async def lifespan(app: FastAPI):
     revival_task.cancel()
     # use the *current* objects — /admin/reload may have swapped them
     for b in state.get("batchers", batchers):
+        await b.close()
+    for b in state.get("dense_batchers") or []:
         await b.close()
     if cfg.drift_state_path:
         state.get("drift_sync", drift_sync).save_state(cfg.drift_state_path)
@@ -459,37 +495,86 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
             )
         return state["metrics"].snapshot()
 
-    @app.post("/predict_dense")
-    async def predict_dense(request: Request):
-        """Dense wide-tabular scoring (BASELINE config 5). Body: binary
-        little-endian f32 — 8-byte header (uint32 rows, uint32 cols) then
-        rows*cols values. Enabled via --dense-model-dir."""
+    def _parse_dense_body(body: bytes, content_type: str, n_features: int):
+        """Two wire formats: binary little-endian f32 (8-byte uint32 rows,
+        uint32 cols header — the bulk path) and JSON {"rows": [[...], ...]}
+        (any standard client; nulls become NaN -> median-imputed)."""
         import struct
 
-        eng = state.get("dense_engine")
-        if eng is None:
-            raise HTTPException(status_code=404, detail="no dense model configured")
-        body = await request.body()
+        if content_type.startswith("application/json") or (
+            body[:1] in (b"{", b"[")
+        ):
+            try:
+                doc = json.loads(body)
+            except ValueError:
+                raise HTTPException(status_code=422, detail="invalid JSON body")
+            rows = doc.get("rows") if isinstance(doc, dict) else doc
+            if not isinstance(rows, list) or not rows:
+                raise HTTPException(
+                    status_code=422, detail='expected {"rows": [[...], ...]}'
+                )
+            try:
+                x = np.array(
+                    [[np.nan if v is None else v for v in r] for r in rows],
+                    dtype=np.float32,
+                )
+            except (TypeError, ValueError) as e:
+                raise HTTPException(status_code=422, detail=f"bad row: {e}")
+            if x.ndim != 2 or x.shape[1] != n_features:
+                raise HTTPException(
+                    status_code=422,
+                    detail=f"expected {n_features} features per row",
+                )
+            return x
         if len(body) < 8:
             raise HTTPException(status_code=422, detail="missing rows/cols header")
         rows, cols = struct.unpack("<II", body[:8])
-        if cols != eng.model.n_features:
+        if cols != n_features:
             raise HTTPException(
                 status_code=422,
-                detail=f"expected {eng.model.n_features} features, got {cols}",
+                detail=f"expected {n_features} features, got {cols}",
             )
         expect = 8 + rows * cols * 4
         if rows == 0 or len(body) != expect:
             raise HTTPException(status_code=422, detail="body size mismatch")
-        x = np.frombuffer(body, dtype="<f4", offset=8).reshape(rows, cols)
-        import asyncio
+        return np.frombuffer(body, dtype="<f4", offset=8).reshape(rows, cols)
 
-        t0 = time.perf_counter()
-        out = await asyncio.get_running_loop().run_in_executor(
-            None, eng.score_arrays, x
+    @app.post("/predict_dense")
+    async def predict_dense(request: Request):
+        """Dense wide-tabular scoring (BASELINE config 5), first-class:
+        micro-batched, dispatched round-robin across the per-GPU dense
+        replicas (each holds the full sorted drift reference in its own
+        HBM), with the same failover semantics as /score."""
+        engines = state.get("dense_engines") or []
+        if not engines:
+            raise HTTPException(status_code=404, detail="no dense model configured")
+        body = await request.body()
+        x = _parse_dense_body(
+            body, request.headers.get("content-type", ""), engines[0].model.n_features
         )
+        pool: ReplicaPool = state["dense_pool"]
+        batchers = state["dense_batchers"]
+        t0 = time.perf_counter()
+        out = None
+        last_exc = None
+        for _ in range(pool.attempt_budget):
+            try:
+                idx = pool.pick()
+            except RuntimeError:
+                state["metrics"].observe_error()
+                raise HTTPException(status_code=503, detail="no healthy dense replicas")
+            try:
+                out = await batchers[idx].submit(x, None)
+                pool.report_ok(idx)
+                break
+            except Exception as e:
+                state["metrics"].observe_error()
+                pool.report_fail(idx)
+                last_exc = e
+        if out is None:
+            raise HTTPException(status_code=500, detail=f"scoring failed: {last_exc}")
         latency_ms = (time.perf_counter() - t0) * 1e3
-        state["metrics"].observe_request(rows, latency_ms)
+        state["metrics"].observe_request(len(x), latency_ms)
         return _json_response(
             {
                 "predictions": np.asarray(out["predictions"]).tolist(),
