@@ -165,10 +165,22 @@ class ScoringEngine:
         recs = json.loads(body)
         if not isinstance(recs, list) or not all(isinstance(r, dict) for r in recs):
             raise ValueError("body must be a JSON array of records")
-        # absent fields take the schema defaults — pydantic default
-        # semantics (reference app/model.py:8-34), matching the native
-        # parser's default-row fill (encode_batch alone would fill missing
-        # categoricals with MISSING_CATEGORY instead)
+        # Strictness parity with the native parser: nulls and ill-typed
+        # values raise ValueError so callers fall back to pydantic for the
+        # reference's 422/coercion semantics; absent fields take the schema
+        # defaults (pydantic default semantics, reference app/model.py:8-34
+        # — encode_batch alone would fill missing categoricals with
+        # MISSING_CATEGORY instead).
+        cat = set(CATEGORICAL_FEATURES)
+        num = set(NUMERIC_FEATURES)
+        for r in recs:
+            for k, v in r.items():
+                if k in cat:
+                    if not isinstance(v, str):
+                        raise ValueError(f"field {k}: expected string")
+                elif k in num:
+                    if isinstance(v, bool) or not isinstance(v, (int, float)):
+                        raise ValueError(f"field {k}: expected number")
         from .schema import LoanApplicant
 
         defaults = LoanApplicant().__dict__

@@ -312,21 +312,29 @@ class RawScoreServer:
     async def _handle(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
         try:
             while True:
-                line = await reader.readline()
-                if not line or line == b"\r\n":
-                    break
+                # one awaited read for request line + all headers (readline
+                # per header cost ~6 event-loop awaits per request)
                 try:
-                    method, path, _ = line.split(b" ", 2)
+                    head = await reader.readuntil(b"\r\n\r\n")
+                except asyncio.LimitOverrunError:
+                    await self._respond(writer, 431, b'{"detail": "headers too large"}')
+                    return
+                except asyncio.IncompleteReadError as e:
+                    if e.partial:  # garbage without a header terminator
+                        break
+                    break  # clean EOF between keep-alive requests
+                lines = head.split(b"\r\n")
+                try:
+                    method, path, _ = lines[0].split(b" ", 2)
                 except ValueError:
                     break
                 clen = 0
                 keep_alive = True
                 chunked = False
                 admin_hdrs = {}
-                while True:
-                    h = await reader.readline()
-                    if h in (b"\r\n", b"\n", b""):
-                        break
+                for h in lines[1:]:
+                    if not h:
+                        continue
                     k, _, v = h.partition(b":")
                     lk = k.lower()
                     if lk == b"content-length":
@@ -410,7 +418,7 @@ class RawScoreServer:
     _REASONS = {200: b"OK", 400: b"Bad Request", 403: b"Forbidden", 404: b"Not Found",
                 405: b"Method Not Allowed", 411: b"Length Required",
                 413: b"Payload Too Large", 422: b"Unprocessable Entity",
-                501: b"Not Implemented",
+                431: b"Request Header Fields Too Large", 501: b"Not Implemented",
                 500: b"Internal Server Error", 503: b"Service Unavailable"}
 
     async def _respond(self, writer, status: int, payload: bytes,
