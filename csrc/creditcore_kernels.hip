@@ -952,8 +952,9 @@ struct ScoreSession {
   int64_t total_bins{}, t_cls{}, t_if{}, capacity{};
   int device_index{};
   hipStream_t stream{};
-  hipStream_t stream2{};  // drift branch (runs parallel to the forests)
-  hipEvent_t ev_fork{}, ev_join{};
+  hipStream_t stream2{};  // K-S branch (runs parallel to the forests)
+  hipStream_t stream3{};  // iforest + categorical-histogram branch
+  hipEvent_t ev_fork{}, ev_join{}, ev_if{}, ev_hist{};
   hipEvent_t ev_done[2]{};  // per-slot completion (async score)
 
   // raw slot pointers into the pinned buffers
@@ -1034,8 +1035,11 @@ struct ScoreSession {
 
     HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&stream2, hipStreamNonBlocking));
+    HIP_CHECK(hipStreamCreateWithFlags(&stream3, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_join, hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&ev_if, hipEventDisableTiming));
+    HIP_CHECK(hipEventCreateWithFlags(&ev_hist, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_done[0], hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_done[1], hipEventDisableTiming));
     HIP_CHECK(hipDeviceSynchronize());  // uploads above used torch's stream
@@ -1047,8 +1051,11 @@ struct ScoreSession {
       for (auto& kv : graphs) (void)hipGraphExecDestroy(kv.second);
       (void)hipStreamDestroy(stream);
       (void)hipStreamDestroy(stream2);
+      (void)hipStreamDestroy(stream3);
       (void)hipEventDestroy(ev_fork);
       (void)hipEventDestroy(ev_join);
+      (void)hipEventDestroy(ev_if);
+      (void)hipEventDestroy(ev_hist);
       (void)hipEventDestroy(ev_done[0]);
       (void)hipEventDestroy(ev_done[1]);
     }
@@ -1057,20 +1064,32 @@ struct ScoreSession {
   // Record the full scoring sequence for batch size b on `stream`.
   // Output layout (b-packed so one D2H covers all three): outs holds
   // proba[0:b] | iscore[b:2b] | outlier[2b:3b]; pin_outs mirrors it.
+  //
+  // Forked shape (b > 64, with drift) — three parallel branches off the
+  // H2D copies, chosen so the classifier forest (the longest kernel,
+  // ~33 µs at b=1024) bounds the graph instead of the serial sum:
+  //   stream : cls forest ─────────────┐
+  //   stream3: iforest → cat_hist      ├ finalize (waits iforest) → outs
+  //   stream2: K-S ── (waits hist) ── drift D2H                     D2H
+  // Round-1 ran iforest serially after the classifier on `stream`
+  // (~9.4 µs of avoidable critical path) and the histogram ahead of K-S
+  // on stream2; measured A/B for this shape is in kernel_tuning.md.
   void record(int b, bool with_drift, int slot) {
     HIP_CHECK(hipMemcpyAsync(d_codes.data_ptr(), p_codes(slot),
         (size_t)b * N_CAT * sizeof(short), hipMemcpyHostToDevice, stream));
     HIP_CHECK(hipMemcpyAsync(d_nums.data_ptr(), p_nums(slot),
         (size_t)b * N_NUM * sizeof(float), hipMemcpyHostToDevice, stream));
-    // fork point: the drift branch (stream2) depends only on the H2D
-    // copies. At tiny batches the drift kernels are microseconds and the
-    // fork/join event edges cost more graph-replay overhead than the
-    // overlap saves — run the branch inline on `stream` instead.
-    const bool fork_drift = with_drift && b > 64;
-    hipStream_t sdrift = fork_drift ? stream2 : stream;
-    if (fork_drift) {
+    // fork point: the side branches depend only on the H2D copies. At tiny
+    // batches the kernels are microseconds and the fork/join event edges
+    // cost more graph-replay overhead than the overlap saves — run
+    // everything serially on `stream` instead.
+    const bool forked = with_drift && b > 64;
+    hipStream_t sdrift = forked ? stream2 : stream;  // K-S + drift D2H
+    hipStream_t sif = forked ? stream3 : stream;     // iforest + histogram
+    if (forked) {
       HIP_CHECK(hipEventRecord(ev_fork, stream));
       HIP_CHECK(hipStreamWaitEvent(stream2, ev_fork, 0));
+      HIP_CHECK(hipStreamWaitEvent(stream3, ev_fork, 0));
     }
     double* acc_cls = acc.data_ptr<double>();
     double* acc_if = acc_cls + b;  // b-packed; all-zero by invariant
@@ -1082,38 +1101,30 @@ struct ScoreSession {
     auto chunks = [&](int64_t t) {
       return std::max(1, std::min(ceil_div(2048, row_blocks), (int)((t + 1) / 2)));
     };
+    hipLaunchKernelGGL((forest_kernel_ilp<true>), dim3(row_blocks, chunks(t_if)),
+        dim3(BLOCK), 0, sif,
+        d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
+        reinterpret_cast<const int4*>(if_nodes.data_ptr<int>()),
+        if_off.data_ptr<int>(), (int)t_if, nullptr, nullptr, b, acc_if);
+    if (forked) HIP_CHECK(hipEventRecord(ev_if, sif));
     hipLaunchKernelGGL((forest_kernel_ilp<false>), dim3(row_blocks, chunks(t_cls)),
         dim3(BLOCK), 0, stream,
         d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
         reinterpret_cast<const int4*>(cls_nodes.data_ptr<int>()),
         cls_off.data_ptr<int>(), (int)t_cls,
         feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), b, acc_cls);
-    hipLaunchKernelGGL((forest_kernel_ilp<true>), dim3(row_blocks, chunks(t_if)),
-        dim3(BLOCK), 0, stream,
-        d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
-        reinterpret_cast<const int4*>(if_nodes.data_ptr<int>()),
-        if_off.data_ptr<int>(), (int)t_if, nullptr, nullptr, b, acc_if);
-
-    double* proba = outs.data_ptr<double>();
-    hipLaunchKernelGGL(finalize_kernel, dim3(row_blocks), dim3(BLOCK), 0, stream,
-        acc_cls, acc_if, b, cls_kind, 1.0 / (double)t_cls, cls_bias,
-        if_denom, if_offset, if_threshold,
-        proba, proba + b, proba + 2 * b);
 
     if (with_drift) {
-      // Drift branch: when forked (b > 64) the K-S and categorical
-      // histogram run on stream2 in parallel with the forest chain
-      // (captured as parallel graph branches, joined after the output
-      // copy); at tiny batches everything stays serial on `stream`.
       // small batches: one block overwrites the histogram (no memset node);
       // larger ones pre-zero + atomically accumulate across blocks
       const int hist_blocks = (b <= 2048) ? 1 : std::min(row_blocks, 1024);
       if (hist_blocks > 1)
-        HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), sdrift));
+        HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), sif));
       hipLaunchKernelGGL(cat_hist_kernel, dim3(hist_blocks), dim3(BLOCK),
-          (size_t)total_bins * sizeof(int), sdrift,
+          (size_t)total_bins * sizeof(int), sif,
           d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
           d_hist());
+      if (forked) HIP_CHECK(hipEventRecord(ev_hist, sif));
       if (b <= KS_COUNT_MAX_ROWS) {
         // O(B^2) counting path: no sort, no barrier chain (38.6 -> single-
         // digit us at b=1024; see profiles/kernel_tuning.md)
@@ -1136,17 +1147,29 @@ struct ScoreSession {
             m_pow2, /*ref_lds=*/0, ref_sorted.data_ptr<float>(),
             rs_off.data_ptr<int64_t>(), d_ksd());
       }
-      // one D2H for the whole drift branch (hist + K-S D share a blob)
+      // one D2H for the whole drift branch (hist + K-S D share a blob);
+      // it needs both the histogram (stream3) and K-S (stream2) done
+      if (forked) HIP_CHECK(hipStreamWaitEvent(sdrift, ev_hist, 0));
       HIP_CHECK(hipMemcpyAsync(p_drift(slot), d_drift.data_ptr<uint8_t>(),
           drift_bytes, hipMemcpyDeviceToHost, sdrift));
-      if (fork_drift) HIP_CHECK(hipEventRecord(ev_join, sdrift));
+      if (forked) HIP_CHECK(hipEventRecord(ev_join, sdrift));
     }
+
+    // finalize consumes both accumulators: when forked it waits on the
+    // iforest branch (done long before the classifier forest it follows)
+    if (forked) HIP_CHECK(hipStreamWaitEvent(stream, ev_if, 0));
+    double* proba = outs.data_ptr<double>();
+    hipLaunchKernelGGL(finalize_kernel, dim3(row_blocks), dim3(BLOCK), 0, stream,
+        acc_cls, acc_if, b, cls_kind, 1.0 / (double)t_cls, cls_bias,
+        if_denom, if_offset, if_threshold,
+        proba, proba + b, proba + 2 * b);
+
     // classifier-output D2H depends only on finalize — it overlaps the
     // drift branch's K-S tail; the join lands after it so graph completion
-    // still covers both streams
+    // still covers both side streams (ev_join transitively orders ev_hist)
     HIP_CHECK(hipMemcpyAsync(p_outs(slot), proba,
         (size_t)(3 * b) * sizeof(double), hipMemcpyDeviceToHost, stream));
-    if (fork_drift) HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
+    if (forked) HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
     HIP_CHECK(hipGetLastError());
   }
 
