@@ -253,3 +253,40 @@ def test_dense_served_concurrent_micro_batching(dense_served):
         np.testing.assert_allclose(
             r.json()["predictions"], ref["predictions"], rtol=1e-6
         )
+
+
+@pytest.mark.gpu
+def test_dense_served_on_gpu(dense_model, model_dir, tmp_path_factory):
+    """/predict_dense through the batcher on the GPU engine: binary and
+    JSON bodies, parity with the CPU reference."""
+    import struct
+
+    from fastapi.testclient import TestClient
+
+    from creditcore.config import ServeConfig
+    from creditcore.serve import create_app
+
+    d = str(tmp_path_factory.mktemp("dense_model_gpu"))
+    dense_model.save(d)
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.dense_model_dir = d
+    cfg.device = "cuda"
+    rng = np.random.default_rng(11)
+    x = rng.normal(size=(64, dense_model.n_features)).astype(np.float32)
+    ref = DenseEngine(dense_model, device="cpu").score_arrays(x)
+    with TestClient(create_app(cfg)) as client:
+        body = struct.pack("<II", *x.shape) + x.tobytes()
+        r = client.post(
+            "/predict_dense", content=body,
+            headers={"Content-Type": "application/octet-stream"},
+        )
+        assert r.status_code == 200, r.text
+        np.testing.assert_allclose(
+            r.json()["predictions"], ref["predictions"], rtol=1e-4, atol=1e-6
+        )
+        r2 = client.post("/predict_dense", json={"rows": x[:8].tolist()})
+        assert r2.status_code == 200
+        np.testing.assert_allclose(
+            r2.json()["predictions"], ref["predictions"][:8], rtol=1e-4, atol=1e-6
+        )
