@@ -76,3 +76,22 @@ def test_config_env_and_cli_precedence(monkeypatch):
     assert cfg.model_directory == "/env/model"
     cfg2 = ServeConfig.from_args(["--port", "6000"])
     assert cfg2.port == 6000  # CLI wins over env default
+
+
+def test_compose_deploy_target_is_valid():
+    """Second deploy target (the ACA analog, reference
+    deploy-container-app.yml): compose file parses and carries the
+    staging->production gate + health checks."""
+    import os
+
+    import yaml
+
+    p = os.path.join(os.path.dirname(os.path.dirname(__file__)),
+                     "deploy", "docker-compose.yml")
+    doc = yaml.safe_load(open(p))
+    svcs = doc["services"]
+    assert {"staging", "production"} <= set(svcs)
+    assert svcs["production"]["depends_on"]["staging"]["condition"] == "service_healthy"
+    for s in ("staging", "production"):
+        assert "healthcheck" in svcs[s]
+        assert any("5000" in p for p in svcs[s]["ports"])
