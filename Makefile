@@ -1,5 +1,9 @@
 # creditcore — common entry points (see docs/runbook.md)
-.PHONY: build test test-gpu bench serve train smoke pipeline clean
+
+lint:
+	python tools/lint.py
+
+.PHONY: lint build test test-gpu bench serve train smoke pipeline clean
 
 build:
 	python setup.py build_ext --inplace

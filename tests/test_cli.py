@@ -95,3 +95,18 @@ def test_compose_deploy_target_is_valid():
     for s in ("staging", "production"):
         assert "healthcheck" in svcs[s]
         assert any("5000" in p for p in svcs[s]["ports"])
+
+
+def test_lint_gate_clean():
+    """Static-check analog of the reference's bicep lint + dependabot
+    (SURVEY §4 static checks): AST lint + dependency-pin freshness."""
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(__file__))
+    r = subprocess.run(
+        [sys.executable, os.path.join(repo, "tools", "lint.py")],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode == 0, r.stdout + r.stderr
