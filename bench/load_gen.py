@@ -25,11 +25,12 @@ import sys as _sys
 _sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
 
 
-async def worker(client, url, body, stop_at, stats):
+async def worker(client, url, body, stop_at, stats, endpoint="/score",
+                 ctype="application/json"):
     while time.perf_counter() < stop_at:
         t0 = time.perf_counter()
-        r = await client.post(f"{url}/score", content=body,
-                              headers={"content-type": "application/json"})
+        r = await client.post(f"{url}{endpoint}", content=body,
+                              headers={"content-type": ctype})
         dt = (time.perf_counter() - t0) * 1e3
         if r.status_code == 200:
             stats["lat"].append(dt)
@@ -37,14 +38,15 @@ async def worker(client, url, body, stop_at, stats):
             stats["errors"] += 1
 
 
-async def raw_worker(host, port, body, stop_at, stats):
+async def raw_worker(host, port, body, stop_at, stats, endpoint=b"/score",
+                     ctype=b"application/json"):
     """Minimal HTTP/1.1 keep-alive client on a raw socket: the httpx stack
     costs ~1.4 ms/request of client CPU, which caps what the *server* can
     be measured at; this one is ~50 µs/request."""
     reader, writer = await asyncio.open_connection(host, port)
     req = (
-        b"POST /score HTTP/1.1\r\nHost: l\r\nContent-Type: application/json\r\n"
-        b"Content-Length: " + str(len(body)).encode() + b"\r\n\r\n" + body
+        b"POST " + endpoint + b" HTTP/1.1\r\nHost: l\r\nContent-Type: " + ctype
+        + b"\r\nContent-Length: " + str(len(body)).encode() + b"\r\n\r\n" + body
     )
     try:
         while time.perf_counter() < stop_at:
@@ -72,7 +74,15 @@ async def run(args):
 
     from creditcore.data import make_request_batch
 
-    body = json.dumps(make_request_batch(args.rows, seed=1)).encode()
+    if getattr(args, "dense_features", 0):
+        # binary dense body for /predict_dense (BASELINE config 5 served)
+        import struct
+
+        rng = np.random.default_rng(1)
+        x = rng.normal(size=(args.rows, args.dense_features)).astype("<f4")
+        body = struct.pack("<II", args.rows, args.dense_features) + x.tobytes()
+    else:
+        body = json.dumps(make_request_batch(args.rows, seed=1)).encode()
     stats = {"lat": [], "errors": 0}
     if getattr(args, "raw_client", False):
         from urllib.parse import urlparse
@@ -81,23 +91,28 @@ async def run(args):
         host, port = u.hostname, u.port or 80
         # warmup (also verifies the server answers)
         w = {"lat": [], "errors": 0}
-        await raw_worker(host, port, body, time.perf_counter() + 0.5, w)
+        ep = args.endpoint.encode()
+        ct = (b"application/octet-stream" if getattr(args, "dense_features", 0)
+              else b"application/json")
+        await raw_worker(host, port, body, time.perf_counter() + 0.5, w, ep, ct)
         stop_at = time.perf_counter() + args.duration
         t0 = time.perf_counter()
         await asyncio.gather(
-            *(raw_worker(host, port, body, stop_at, stats)
+            *(raw_worker(host, port, body, stop_at, stats, ep, ct)
               for _ in range(args.concurrency))
         )
         elapsed = time.perf_counter() - t0
     else:
       async with httpx.AsyncClient(timeout=60.0) as client:
         # warmup
-        await client.post(f"{args.url}/score", content=body,
-                          headers={"content-type": "application/json"})
+        ct = ("application/octet-stream" if getattr(args, "dense_features", 0)
+              else "application/json")
+        await client.post(f"{args.url}{args.endpoint}", content=body,
+                          headers={"content-type": ct})
         stop_at = time.perf_counter() + args.duration
         t0 = time.perf_counter()
         await asyncio.gather(
-            *(worker(client, args.url, body, stop_at, stats)
+            *(worker(client, args.url, body, stop_at, stats, args.endpoint, ct)
               for _ in range(args.concurrency))
         )
         elapsed = time.perf_counter() - t0
@@ -138,6 +153,10 @@ def main():
     p.add_argument("--processes", type=int, default=1,
                    help="client processes (a single event loop saturates "
                         "before the server does on large bodies)")
+    p.add_argument("--endpoint", default="/score")
+    p.add_argument("--dense-features", type=int, default=0,
+                   help="send binary dense bodies with this many features "
+                        "(use with --endpoint /predict_dense)")
     p.add_argument("--raw-client", action="store_true",
                    help="raw-socket HTTP client (httpx costs ~1.4 ms/req "
                         "of client CPU and caps the measurement)")
