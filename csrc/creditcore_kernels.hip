@@ -184,6 +184,117 @@ __global__ __launch_bounds__(BLOCK) void forest_kernel_ilp(
   atomicAdd(&acc[row], local);
 }
 
+// Dual-forest fusion: classifier (one-hot resolution) and isolation forest
+// (direct numeric features) in ONE launch — grid-y is split into
+// chunks_cls classifier chunks followed by iforest chunks, each block
+// running the same 2-tree-ILP walk against its forest's tables. The two
+// forests were two serial launches in round 1 (~9.4 µs of iforest after
+// ~33 µs of classifier at b=1024); one grid lets the scheduler run both
+// concurrently with NO new graph edges (a 3-stream fork was measured
+// SLOWER — the extra cross-stream edges cost ~20 µs/replay, see
+// kernel_tuning.md).
+template <bool DIRECT>
+__device__ __forceinline__ double forest_walk_ilp2(
+    const short* __restrict__ s_codes,
+    const float* __restrict__ s_nums,
+    int tid,
+    const int4* __restrict__ nodes,
+    const int* __restrict__ tree_off,
+    int n_trees,
+    const int* __restrict__ feat_col,
+    const int* __restrict__ feat_code,
+    int chunk,
+    int n_chunks)
+{
+  auto value_of = [&](int f) -> float {
+    if (DIRECT) return s_nums[f * BLOCK + tid];
+    const int col = feat_col[f];
+    const int code = feat_code[f];
+    return (code >= 0) ? ((s_codes[col * BLOCK + tid] == (short)code) ? 1.0f : 0.0f)
+                       : s_nums[col * BLOCK + tid];
+  };
+  double local = 0.0;
+  for (int t = chunk; t < n_trees; t += 2 * n_chunks) {
+    const int ta = t;
+    const int tb = t + n_chunks;
+    const int basea = tree_off[ta];
+    int4 na = nodes[basea];
+    bool la = true;
+    int baseb = 0;
+    int4 nb;
+    bool lb = tb < n_trees;
+    if (lb) { baseb = tree_off[tb]; nb = nodes[baseb]; }
+    while (la || lb) {
+      if (la) {
+        if (na.x >= 0) {
+          const float v = value_of(na.x);
+          na = nodes[basea + ((v <= __int_as_float(na.y)) ? na.z : na.w)];
+        } else {
+          local += (double)__int_as_float(na.y);
+          la = false;
+        }
+      }
+      if (lb) {
+        if (nb.x >= 0) {
+          const float v = value_of(nb.x);
+          nb = nodes[baseb + ((v <= __int_as_float(nb.y)) ? nb.z : nb.w)];
+        } else {
+          local += (double)__int_as_float(nb.y);
+          lb = false;
+        }
+      }
+    }
+  }
+  return local;
+}
+
+__global__ __launch_bounds__(BLOCK) void forest_kernel_dual(
+    const short* __restrict__ codes,
+    const float* __restrict__ nums,
+    const float* __restrict__ medians,
+    const int4* __restrict__ cls_nodes,
+    const int* __restrict__ cls_off,
+    int t_cls,
+    const int* __restrict__ feat_col,
+    const int* __restrict__ feat_code,
+    const int4* __restrict__ if_nodes,
+    const int* __restrict__ if_off,
+    int t_if,
+    int chunks_cls,
+    int n_rows,
+    double* __restrict__ acc_cls,
+    double* __restrict__ acc_if)
+{
+  __shared__ short s_codes[N_CAT * BLOCK];
+  __shared__ float s_nums[N_NUM * BLOCK];
+  const int tid = threadIdx.x;
+  const int row = blockIdx.x * BLOCK + tid;
+  const bool is_cls = (int)blockIdx.y < chunks_cls;
+  if (row < n_rows) {
+    if (is_cls) {
+#pragma unroll
+      for (int c = 0; c < N_CAT; ++c) s_codes[c * BLOCK + tid] = codes[row * N_CAT + c];
+    }
+#pragma unroll
+    for (int c = 0; c < N_NUM; ++c) {
+      const float v = nums[row * N_NUM + c];
+      s_nums[c * BLOCK + tid] = isnan(v) ? medians[c] : v;
+    }
+  }
+  if (row >= n_rows) return;
+  if (is_cls) {
+    const double local = forest_walk_ilp2<false>(
+        s_codes, s_nums, tid, cls_nodes, cls_off, t_cls, feat_col, feat_code,
+        blockIdx.y, chunks_cls);
+    atomicAdd(&acc_cls[row], local);
+  } else {
+    const double local = forest_walk_ilp2<true>(
+        s_codes, s_nums, tid, if_nodes, if_off, t_if, nullptr, nullptr,
+        blockIdx.y - chunks_cls, gridDim.y - chunks_cls);
+    atomicAdd(&acc_if[row], local);
+  }
+}
+
 // 4-tree ILP + optional transposed grid (blockIdx.x = tree chunk so the
 // dispatcher's id%8 XCD placement gives adjacent chunks to different XCDs,
 // keeping each XCD's L2 on a tree subset). Experimental A/B variants.
@@ -952,9 +1063,8 @@ struct ScoreSession {
   int64_t total_bins{}, t_cls{}, t_if{}, capacity{};
   int device_index{};
   hipStream_t stream{};
-  hipStream_t stream2{};  // K-S branch (runs parallel to the forests)
-  hipStream_t stream3{};  // iforest + categorical-histogram branch
-  hipEvent_t ev_fork{}, ev_join{}, ev_if{}, ev_hist{};
+  hipStream_t stream2{};  // drift branch (runs parallel to the forests)
+  hipEvent_t ev_fork{}, ev_join{};
   hipEvent_t ev_done[2]{};  // per-slot completion (async score)
 
   // raw slot pointers into the pinned buffers
@@ -1035,11 +1145,8 @@ struct ScoreSession {
 
     HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
     HIP_CHECK(hipStreamCreateWithFlags(&stream2, hipStreamNonBlocking));
-    HIP_CHECK(hipStreamCreateWithFlags(&stream3, hipStreamNonBlocking));
     HIP_CHECK(hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_join, hipEventDisableTiming));
-    HIP_CHECK(hipEventCreateWithFlags(&ev_if, hipEventDisableTiming));
-    HIP_CHECK(hipEventCreateWithFlags(&ev_hist, hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_done[0], hipEventDisableTiming));
     HIP_CHECK(hipEventCreateWithFlags(&ev_done[1], hipEventDisableTiming));
     HIP_CHECK(hipDeviceSynchronize());  // uploads above used torch's stream
@@ -1051,11 +1158,8 @@ struct ScoreSession {
       for (auto& kv : graphs) (void)hipGraphExecDestroy(kv.second);
       (void)hipStreamDestroy(stream);
       (void)hipStreamDestroy(stream2);
-      (void)hipStreamDestroy(stream3);
       (void)hipEventDestroy(ev_fork);
       (void)hipEventDestroy(ev_join);
-      (void)hipEventDestroy(ev_if);
-      (void)hipEventDestroy(ev_hist);
       (void)hipEventDestroy(ev_done[0]);
       (void)hipEventDestroy(ev_done[1]);
     }
@@ -1065,31 +1169,27 @@ struct ScoreSession {
   // Output layout (b-packed so one D2H covers all three): outs holds
   // proba[0:b] | iscore[b:2b] | outlier[2b:3b]; pin_outs mirrors it.
   //
-  // Forked shape (b > 64, with drift) — three parallel branches off the
-  // H2D copies, chosen so the classifier forest (the longest kernel,
-  // ~33 µs at b=1024) bounds the graph instead of the serial sum:
-  //   stream : cls forest ─────────────┐
-  //   stream3: iforest → cat_hist      ├ finalize (waits iforest) → outs
-  //   stream2: K-S ── (waits hist) ── drift D2H                     D2H
-  // Round-1 ran iforest serially after the classifier on `stream`
-  // (~9.4 µs of avoidable critical path) and the histogram ahead of K-S
-  // on stream2; measured A/B for this shape is in kernel_tuning.md.
+  // Two-stream shape (round-1 tuned): the drift branch (histogram + K-S)
+  // forks onto stream2 off the H2D copies and joins after the output
+  // copy. Both forests run as ONE dual-grid launch (forest_kernel_dual)
+  // so the iforest rides along with the classifier instead of
+  // serializing after it; a 3-stream fork for the same effect was
+  // measured SLOWER (extra cross-stream graph edges ≈ +20 µs/replay at
+  // b=1024 — kernel_tuning.md).
   void record(int b, bool with_drift, int slot) {
     HIP_CHECK(hipMemcpyAsync(d_codes.data_ptr(), p_codes(slot),
         (size_t)b * N_CAT * sizeof(short), hipMemcpyHostToDevice, stream));
     HIP_CHECK(hipMemcpyAsync(d_nums.data_ptr(), p_nums(slot),
         (size_t)b * N_NUM * sizeof(float), hipMemcpyHostToDevice, stream));
-    // fork point: the side branches depend only on the H2D copies. At tiny
-    // batches the kernels are microseconds and the fork/join event edges
-    // cost more graph-replay overhead than the overlap saves — run
-    // everything serially on `stream` instead.
-    const bool forked = with_drift && b > 64;
-    hipStream_t sdrift = forked ? stream2 : stream;  // K-S + drift D2H
-    hipStream_t sif = forked ? stream3 : stream;     // iforest + histogram
-    if (forked) {
+    // fork point: the drift branch (stream2) depends only on the H2D
+    // copies. At tiny batches the drift kernels are microseconds and the
+    // fork/join event edges cost more graph-replay overhead than the
+    // overlap saves — run the branch inline on `stream` instead.
+    const bool fork_drift = with_drift && b > 64;
+    hipStream_t sdrift = fork_drift ? stream2 : stream;
+    if (fork_drift) {
       HIP_CHECK(hipEventRecord(ev_fork, stream));
       HIP_CHECK(hipStreamWaitEvent(stream2, ev_fork, 0));
-      HIP_CHECK(hipStreamWaitEvent(stream3, ev_fork, 0));
     }
     double* acc_cls = acc.data_ptr<double>();
     double* acc_if = acc_cls + b;  // b-packed; all-zero by invariant
@@ -1097,34 +1197,41 @@ struct ScoreSession {
     const int row_blocks = ceil_div(b, BLOCK);
     // 2-tree-ILP traversal measured faster at every batch size
     // (bench/kernel_micro.py: -12% @1k rows, -42% @16k); grid-y covers
-    // ceil(T/2) chunks, each thread walking trees t and t+gridDim.y.
+    // ceil(T/2) chunks, each thread walking trees t and t+n_chunks.
     auto chunks = [&](int64_t t) {
       return std::max(1, std::min(ceil_div(2048, row_blocks), (int)((t + 1) / 2)));
     };
-    hipLaunchKernelGGL((forest_kernel_ilp<true>), dim3(row_blocks, chunks(t_if)),
-        dim3(BLOCK), 0, sif,
-        d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
-        reinterpret_cast<const int4*>(if_nodes.data_ptr<int>()),
-        if_off.data_ptr<int>(), (int)t_if, nullptr, nullptr, b, acc_if);
-    if (forked) HIP_CHECK(hipEventRecord(ev_if, sif));
-    hipLaunchKernelGGL((forest_kernel_ilp<false>), dim3(row_blocks, chunks(t_cls)),
+    const int c_cls = chunks(t_cls);
+    const int c_if = chunks(t_if);
+    hipLaunchKernelGGL(forest_kernel_dual, dim3(row_blocks, c_cls + c_if),
         dim3(BLOCK), 0, stream,
         d_codes.data_ptr<short>(), d_nums.data_ptr<float>(), medians.data_ptr<float>(),
         reinterpret_cast<const int4*>(cls_nodes.data_ptr<int>()),
         cls_off.data_ptr<int>(), (int)t_cls,
-        feat_col.data_ptr<int>(), feat_code.data_ptr<int>(), b, acc_cls);
+        feat_col.data_ptr<int>(), feat_code.data_ptr<int>(),
+        reinterpret_cast<const int4*>(if_nodes.data_ptr<int>()),
+        if_off.data_ptr<int>(), (int)t_if, c_cls, b, acc_cls, acc_if);
+
+    double* proba = outs.data_ptr<double>();
+    hipLaunchKernelGGL(finalize_kernel, dim3(row_blocks), dim3(BLOCK), 0, stream,
+        acc_cls, acc_if, b, cls_kind, 1.0 / (double)t_cls, cls_bias,
+        if_denom, if_offset, if_threshold,
+        proba, proba + b, proba + 2 * b);
 
     if (with_drift) {
+      // Drift branch: when forked (b > 64) the K-S and categorical
+      // histogram run on stream2 in parallel with the forest chain
+      // (captured as parallel graph branches, joined after the output
+      // copy); at tiny batches everything stays serial on `stream`.
       // small batches: one block overwrites the histogram (no memset node);
       // larger ones pre-zero + atomically accumulate across blocks
       const int hist_blocks = (b <= 2048) ? 1 : std::min(row_blocks, 1024);
       if (hist_blocks > 1)
-        HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), sif));
+        HIP_CHECK(hipMemsetAsync(d_hist(), 0, (size_t)total_bins * sizeof(int), sdrift));
       hipLaunchKernelGGL(cat_hist_kernel, dim3(hist_blocks), dim3(BLOCK),
-          (size_t)total_bins * sizeof(int), sif,
+          (size_t)total_bins * sizeof(int), sdrift,
           d_codes.data_ptr<short>(), b, cat_off.data_ptr<int>(), (int)total_bins,
           d_hist());
-      if (forked) HIP_CHECK(hipEventRecord(ev_hist, sif));
       if (b <= KS_COUNT_MAX_ROWS) {
         // O(B^2) counting path: no sort, no barrier chain (38.6 -> single-
         // digit us at b=1024; see profiles/kernel_tuning.md)
@@ -1147,29 +1254,17 @@ struct ScoreSession {
             m_pow2, /*ref_lds=*/0, ref_sorted.data_ptr<float>(),
             rs_off.data_ptr<int64_t>(), d_ksd());
       }
-      // one D2H for the whole drift branch (hist + K-S D share a blob);
-      // it needs both the histogram (stream3) and K-S (stream2) done
-      if (forked) HIP_CHECK(hipStreamWaitEvent(sdrift, ev_hist, 0));
+      // one D2H for the whole drift branch (hist + K-S D share a blob)
       HIP_CHECK(hipMemcpyAsync(p_drift(slot), d_drift.data_ptr<uint8_t>(),
           drift_bytes, hipMemcpyDeviceToHost, sdrift));
-      if (forked) HIP_CHECK(hipEventRecord(ev_join, sdrift));
+      if (fork_drift) HIP_CHECK(hipEventRecord(ev_join, sdrift));
     }
-
-    // finalize consumes both accumulators: when forked it waits on the
-    // iforest branch (done long before the classifier forest it follows)
-    if (forked) HIP_CHECK(hipStreamWaitEvent(stream, ev_if, 0));
-    double* proba = outs.data_ptr<double>();
-    hipLaunchKernelGGL(finalize_kernel, dim3(row_blocks), dim3(BLOCK), 0, stream,
-        acc_cls, acc_if, b, cls_kind, 1.0 / (double)t_cls, cls_bias,
-        if_denom, if_offset, if_threshold,
-        proba, proba + b, proba + 2 * b);
-
     // classifier-output D2H depends only on finalize — it overlaps the
     // drift branch's K-S tail; the join lands after it so graph completion
-    // still covers both side streams (ev_join transitively orders ev_hist)
+    // still covers both streams
     HIP_CHECK(hipMemcpyAsync(p_outs(slot), proba,
         (size_t)(3 * b) * sizeof(double), hipMemcpyDeviceToHost, stream));
-    if (forked) HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
+    if (fork_drift) HIP_CHECK(hipStreamWaitEvent(stream, ev_join, 0));
     HIP_CHECK(hipGetLastError());
   }
 
