@@ -321,9 +321,11 @@ def main():
     if device == "cuda":
         t_burn = time.perf_counter()
         burn = 0
-        while burn < 256 and time.perf_counter() - t_burn < 2.0:
-            run_steps(16)
-            burn += 16
+        # long enough to reach thermal/clock steady state (~0.5 s of
+        # continuous graph replays), still bounded for big models
+        while burn < 4096 and time.perf_counter() - t_burn < 3.0:
+            run_steps(64)
+            burn += 64
         print(f"[bench] burn-in: {burn} steps in "
               f"{time.perf_counter() - t_burn:.2f}s", file=sys.stderr)
     run_steps(args.warmup)
