@@ -645,6 +645,113 @@ __global__ __launch_bounds__(BS) void ks_count_kernel_t(
 // clearly on the small batches the single-call latency path serves
 constexpr int KS_COUNT_MAX_ROWS = 256;
 
+// Multi-block counting K-S: the single-block kernels occupy only n_cols
+// (=14) of the 256 CUs. This variant splits each column's element range
+// over gridDim.y blocks — every block stages the full imputed column in
+// LDS (B ≤ 16384 ⇒ ≤ 64 KiB) and sweeps its slice with the same
+// LDS-broadcast quadratic count, so ~200+ workgroups fill the chip and
+// the O(B²) work parallelizes. Per-(column, block) partial maxima land in
+// `scratch`; ks_count_reduce_kernel folds them into ks_d. lt/le counting
+// is identical to the single-block kernels (sorted-path tie-run bounds),
+// and f64→f32 casting commutes with max, so the result is bitwise equal
+// to ks_kernel_t's.
+template <int BS>
+__global__ __launch_bounds__(BS) void ks_count_mb_kernel_t(
+    const float* __restrict__ nums,       // [B, n_cols]
+    const float* __restrict__ medians,    // [n_cols]
+    int n_cols, int n_rows,
+    const float* __restrict__ ref_sorted, const int64_t* __restrict__ rs_off,
+    float* __restrict__ scratch)          // [n_cols, gridDim.y]
+{
+  extern __shared__ float s_vals[];  // [n_rows] imputed batch column
+  const int j = blockIdx.x;
+  const int m = n_rows;
+  for (int i = threadIdx.x; i < m; i += blockDim.x) {
+    float v = nums[i * n_cols + j];
+    if (isnan(v)) v = medians[j];
+    s_vals[i] = v;
+  }
+  const int64_t ref_lo = rs_off[j];
+  const int n = (int)(rs_off[j + 1] - ref_lo);
+  const float* __restrict__ ref = ref_sorted + ref_lo;
+  __syncthreads();
+
+  // this block's element slice: [e0, e1)
+  const int per = (m + gridDim.y - 1) / gridDim.y;
+  const int e0 = blockIdx.y * per;
+  const int e1 = min(m, e0 + per);
+
+  double dmax = 0.0;
+  for (int i = e0 + threadIdx.x; i < e1; i += blockDim.x) {
+    const float x = s_vals[i];
+    int lt = 0, le = 0;
+    int k = 0;
+    for (; k + 3 < m; k += 4) {
+      const float v0 = s_vals[k], v1 = s_vals[k + 1];
+      const float v2 = s_vals[k + 2], v3 = s_vals[k + 3];
+      lt += (int)(v0 < x) + (int)(v1 < x) + (int)(v2 < x) + (int)(v3 < x);
+      le += (int)(v0 <= x) + (int)(v1 <= x) + (int)(v2 <= x) + (int)(v3 <= x);
+    }
+    for (; k < m; ++k) {
+      const float v = s_vals[k];
+      lt += (int)(v < x);
+      le += (int)(v <= x);
+    }
+    int l = 0, r = n;
+    while (l < r) {
+      const int mid = (l + r) >> 1;
+      if (ref[mid] < x) l = mid + 1; else r = mid;
+    }
+    const int sl = l;
+    r = n;
+    while (l < r) {
+      const int mid = (l + r) >> 1;
+      if (ref[mid] <= x) l = mid + 1; else r = mid;
+    }
+    const int sr = l;
+    const double fl = fabs((double)sl / n - (double)lt / m);
+    const double fr = fabs((double)sr / n - (double)le / m);
+    dmax = fmax(dmax, fmax(fl, fr));
+  }
+
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    dmax = fmax(dmax, __shfl_down(dmax, off, 64));
+  __syncthreads();
+  float* s_red = s_vals;  // one slot per wave (smem floors at BS/64)
+  const int wave = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) s_red[wave] = (float)dmax;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float d = s_red[0];
+    for (int w = 1; w < (int)(blockDim.x >> 6); ++w) d = fmaxf(d, s_red[w]);
+    scratch[(int64_t)j * gridDim.y + blockIdx.y] = d;
+  }
+}
+
+__global__ __launch_bounds__(64) void ks_count_reduce_kernel(
+    const float* __restrict__ scratch,  // [n_cols, g]
+    int g, int n_cols,
+    float* __restrict__ ks_d)
+{
+  const int j = blockIdx.x * 64 + threadIdx.x;
+  if (j >= n_cols) return;
+  float d = 0.0f;
+  for (int k = 0; k < g; ++k) d = fmaxf(d, scratch[(int64_t)j * g + k]);
+  ks_d[j] = d;
+}
+
+// blocks per column for the multi-block counting path: ~128 elements per
+// block keeps every block's sweep a few microseconds while n_cols*g
+// workgroups fill the 256 CUs
+inline int ks_mb_blocks(int b) {
+  int g = (b + 127) / 128;
+  if (g < 1) g = 1;
+  if (g > 128) g = 128;
+  return g;
+}
+constexpr int KS_MB_MAX_G = 128;
+
 // ---------------------------------------------------------------------------
 // Host launchers
 // ---------------------------------------------------------------------------
@@ -1053,7 +1160,7 @@ torch::Tensor ks_stats(
 struct ScoreSession {
   torch::Tensor cls_nodes, cls_off, feat_col, feat_code, medians;
   torch::Tensor if_nodes, if_off, ref_sorted, rs_off, cat_off;
-  torch::Tensor d_codes, d_nums, acc, outs, d_drift;
+  torch::Tensor d_codes, d_nums, acc, outs, d_drift, d_ks_scratch;
   size_t drift_bytes = 0, drift_ksd_off = 0;  // [hist i32 | ksd f32] blob
   torch::Tensor pin_codes, pin_nums, pin_outs, pin_drift;
   torch::Tensor pin_hist, pin_ksd;  // typed views into pin_drift
@@ -1117,6 +1224,9 @@ struct ScoreSession {
     drift_ksd_off = (size_t)total_bins * sizeof(int32_t);
     drift_bytes = drift_ksd_off + (size_t)N_NUM * sizeof(float);
     d_drift = torch::empty({(int64_t)drift_bytes}, devopt.dtype(torch::kUInt8));
+    // per-(column, block) partial maxima for the multi-block counting K-S
+    d_ks_scratch = torch::empty({(int64_t)N_NUM * KS_MB_MAX_G},
+                                devopt.dtype(torch::kFloat32));
 
     // two slots: step i's host epilogue reads slot i%2 while step i+1's
     // graph fills the other slot (pipelined serving/bench loops)
@@ -1141,6 +1251,9 @@ struct ScoreSession {
     (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&ks_kernel_t<256>),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)KS_LDS_BYTES);
     (void)hipFuncSetAttribute(reinterpret_cast<const void*>(&ks_kernel_t<512>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, (int)KS_LDS_BYTES);
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&ks_count_mb_kernel_t<512>),
         hipFuncAttributeMaxDynamicSharedMemorySize, (int)KS_LDS_BYTES);
 
     HIP_CHECK(hipStreamCreateWithFlags(&stream, hipStreamNonBlocking));
@@ -1240,7 +1353,9 @@ struct ScoreSession {
             smem, sdrift,
             d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b,
             ref_sorted.data_ptr<float>(), rs_off.data_ptr<int64_t>(), d_ksd());
-      } else {
+      } else if (std::getenv("CREDITCORE_KS_BITONIC") != nullptr) {
+        // round-1 path kept for A/B: one bitonic-sort block per column
+        // (only n_cols of 256 CUs busy; 38.6 us at b=1024, 509 at 16k)
         int m_pow2 = 512 / 64;  // >= one cross-wave reduction slot per wave
         while (m_pow2 < b) m_pow2 <<= 1;
         // measured (bench/kernel_micro.py on MI355X): staging the ref
@@ -1253,6 +1368,18 @@ struct ScoreSession {
             d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b,
             m_pow2, /*ref_lds=*/0, ref_sorted.data_ptr<float>(),
             rs_off.data_ptr<int64_t>(), d_ksd());
+      } else {
+        // multi-block counting path: n_cols*g workgroups fill the chip
+        // (the sort/count single-block kernels used 14 of 256 CUs)
+        const int g = ks_mb_blocks(b);
+        const size_t smem = (size_t)std::max(b, 512 / 64) * sizeof(float);
+        hipLaunchKernelGGL((ks_count_mb_kernel_t<512>), dim3(N_NUM, g),
+            dim3(512), smem, sdrift,
+            d_nums.data_ptr<float>(), medians.data_ptr<float>(), N_NUM, b,
+            ref_sorted.data_ptr<float>(), rs_off.data_ptr<int64_t>(),
+            d_ks_scratch.data_ptr<float>());
+        hipLaunchKernelGGL(ks_count_reduce_kernel, dim3(1), dim3(64), 0,
+            sdrift, d_ks_scratch.data_ptr<float>(), g, N_NUM, d_ksd());
       }
       // one D2H for the whole drift branch (hist + K-S D share a blob)
       HIP_CHECK(hipMemcpyAsync(p_drift(slot), d_drift.data_ptr<uint8_t>(),
