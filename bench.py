@@ -248,7 +248,7 @@ def main():
             outs = []
             t_prev = time.perf_counter()
             for i in range(k):
-                outs.append(one_step(i))
+                outs.append(one_step(i)["rows"])  # response freed per step
                 if step_times is not None:
                     t_now = time.perf_counter()
                     step_times.append(t_now - t_prev)
@@ -285,7 +285,12 @@ def main():
                     torch.from_numpy(nums).to(drift_dev),
                 )
                 drift_sync.allreduce()
-            outs.append(out)
+            # count the step but drop the response object — a server frees
+            # each response after writing it to the socket; retaining all
+            # of them here grew the heap ~25 KB/step, and the allocator's
+            # periodic fresh-arena mmaps showed up as ~200 µs step-time
+            # spikes every ~10 steps
+            outs.append(out["rows"])
             if step_times is not None:
                 t_now = time.perf_counter()
                 step_times.append(t_now - t_prev)
