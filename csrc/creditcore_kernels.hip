@@ -1157,6 +1157,38 @@ torch::Tensor ks_stats(
 // was ~10x the kernel time). The GIL is released while waiting.
 // ---------------------------------------------------------------------------
 
+// Completion waits: hipStreamSynchronize/hipEventSynchronize block on the
+// runtime's interrupt-driven wakeup, which costs tens of µs of wake
+// latency per request — a large share of the ~60 µs/request host floor at
+// microsecond kernel times. Serving processes own their core, so the
+// default is a polling wait (hipStreamQuery granularity ~1-2 µs);
+// CREDITCORE_BLOCK_SYNC=1 restores blocking waits for A/B or
+// share-the-host deployments.
+inline bool spin_sync_enabled() {
+  static const bool on = (std::getenv("CREDITCORE_BLOCK_SYNC") == nullptr);
+  return on;
+}
+
+inline void sync_stream(hipStream_t s) {
+  if (spin_sync_enabled()) {
+    hipError_t e;
+    while ((e = hipStreamQuery(s)) == hipErrorNotReady) {}
+    if (e != hipSuccess) HIP_CHECK(e);
+  } else {
+    HIP_CHECK(hipStreamSynchronize(s));
+  }
+}
+
+inline void sync_event(hipEvent_t ev) {
+  if (spin_sync_enabled()) {
+    hipError_t e;
+    while ((e = hipEventQuery(ev)) == hipErrorNotReady) {}
+    if (e != hipSuccess) HIP_CHECK(e);
+  } else {
+    HIP_CHECK(hipEventSynchronize(ev));
+  }
+}
+
 struct ScoreSession {
   torch::Tensor cls_nodes, cls_off, feat_col, feat_code, medians;
   torch::Tensor if_nodes, if_off, ref_sorted, rs_off, cat_off;
@@ -1418,7 +1450,7 @@ struct ScoreSession {
     if (it == graphs.end()) {
       if (!graphable || graphs.size() >= 128) {
         record(b, with_drift, slot);
-        if (sync) HIP_CHECK(hipStreamSynchronize(stream));
+        if (sync) sync_stream(stream);
         else HIP_CHECK(hipEventRecord(ev_done[slot], stream));
         return;
       }
@@ -1432,13 +1464,13 @@ struct ScoreSession {
       it = graphs.emplace(key, exec).first;
     }
     HIP_CHECK(hipGraphLaunch(it->second, stream));
-    if (sync) HIP_CHECK(hipStreamSynchronize(stream));
+    if (sync) sync_stream(stream);
     else HIP_CHECK(hipEventRecord(ev_done[slot], stream));
   }
 
   void wait_slot(int64_t slot) {
     py::gil_scoped_release nogil;
-    HIP_CHECK(hipEventSynchronize(ev_done[slot & 1]));
+    sync_event(ev_done[slot & 1]);
   }
 
   std::unordered_map<uint64_t, hipGraphExec_t> graphs;
@@ -2252,7 +2284,7 @@ py::bytes response_epilogue(ScoreSession& s, int64_t slot64, int64_t b64,
   std::string out;
   {
     py::gil_scoped_release nogil;
-    HIP_CHECK(hipEventSynchronize(s.ev_done[slot]));
+    sync_event(s.ev_done[slot]);
     drift_pvals_raw(s.p_hist(slot), s.p_ksd(slot), N_NUM,
                     ref_cat_counts.data(), cat_offsets.data(),
                     (int)cat_offsets.size() - 1, n_ref, nb, pv);
