@@ -55,3 +55,33 @@ def test_fastapi_reload_gated(model_dir):
         from creditcore.schema import SAMPLE_REQUEST
 
         assert client.post("/score", json=SAMPLE_REQUEST).status_code == 200
+
+
+def test_fastapi_reload_rejects_remote_client_without_token(model_dir):
+    """No token configured: a non-loopback client address gets 403 on
+    /admin/reload while /score keeps working (the actual reported hole —
+    0.0.0.0 public listener, any remote client could hot-swap the model)."""
+    import anyio
+    import httpx
+
+    from creditcore.serve import create_app
+
+    cfg = _cfg()  # no token -> loopback-only
+    cfg.model_directory = model_dir
+    cfg.device = "cpu"
+    app = create_app(cfg)
+
+    async def go():
+        async with app.router.lifespan_context(app):
+            transport = httpx.ASGITransport(app, client=("203.0.113.9", 4242))
+            async with httpx.AsyncClient(
+                transport=transport, base_url="http://svc"
+            ) as c:
+                r = await c.post("/admin/reload", json={"model_uri": model_dir})
+                assert r.status_code == 403
+                from creditcore.schema import SAMPLE_REQUEST
+
+                ok = await c.post("/score", json=SAMPLE_REQUEST)
+                assert ok.status_code == 200
+
+    anyio.run(go)
