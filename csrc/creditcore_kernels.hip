@@ -1603,16 +1603,25 @@ struct Parser {
   }
 
   // Parse a JSON string starting at the opening quote; returns the raw
-  // span between quotes and whether it contains escapes.
+  // span between quotes and whether it contains escapes. Fast path: one
+  // SIMD memchr to the closing quote + one over the span for backslashes
+  // (escapes are rare in this schema's values).
   void raw_string(const char*& s, size_t& len, bool& escaped) {
     expect('"', "expected string");
     s = p;
-    escaped = false;
+    const char* q = (const char*)std::memchr(p, '"', (size_t)(end - p));
+    if (q == nullptr) { p = end; fail("unterminated string"); }
+    if (std::memchr(p, '\\', (size_t)(q - p)) == nullptr) {
+      escaped = false;
+      len = (size_t)(q - s);
+      p = q + 1;
+      return;
+    }
+    escaped = true;
     while (p < end) {
       const char c = *p;
       if (c == '"') { len = (size_t)(p - s); ++p; return; }
       if (c == '\\') {
-        escaped = true;
         ++p;
         if (p >= end) break;
       }
@@ -1807,6 +1816,10 @@ static size_t parse_records_nogil(const JsonEncoderState& st, char* data,
           // reference app/model.py:8-34)
           std::memcpy(crow, def_codes, ncat * sizeof(int16_t));
           std::memcpy(nrow, def_nums, nnum * sizeof(float));
+          // clients overwhelmingly emit fields in schema order (pydantic
+          // model dumps, the sample request, pandas records): try the
+          // expected next column with one memcmp before the hash lookup
+          int expect_col = 0;
           if (!P.eat('}')) {
             do {
               const char* ks; size_t kl; bool kesc;
@@ -1818,7 +1831,14 @@ static size_t parse_records_nogil(const JsonEncoderState& st, char* data,
                 ks = kbuf; kl = n;
               }
               P.expect(':', "expected ':'");
-              const int col = st.lookup(ks, kl);
+              int col = -1;
+              if (expect_col < ncat + nnum) {
+                const std::string& exp = st.key_store[expect_col];
+                if (exp.size() == kl && std::memcmp(exp.data(), ks, kl) == 0)
+                  col = expect_col;
+              }
+              if (col < 0) col = st.lookup(ks, kl);
+              expect_col = (col >= 0) ? col + 1 : expect_col;
               if (col < 0) {
                 P.skip_value();  // extra fields ignored (schema extra="ignore")
               } else if (col < ncat) {
