@@ -119,3 +119,25 @@ def test_bench_refuses_to_extrapolate_gpus(tmp_path):
     assert r.returncode == 2, (r.returncode, r.stdout[-500:], r.stderr[-500:])
     assert "refusing to extrapolate" in r.stderr
     assert not [l for l in r.stdout.splitlines() if l.startswith("{")]
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(600)
+def test_bench_torchrun_ws4_gloo(tmp_path):
+    """4-rank rehearsal of the driver's scale shape (gloo): broadcast from
+    rank 0, all ranks step, MAX-over-ranks aggregate — the same code path
+    RCCL takes on the 8-GPU node."""
+    port = _free_port()
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", str(port),
+         os.path.join(REPO, "bench.py"), "--device", "cpu", "--gpus", "4",
+         "--steps", "2", "--warmup", "1", "--rows", "32",
+         "--model-trees", "10", "--model-depth", "3"],
+        capture_output=True, text=True, timeout=540, env=_env(tmp_path), cwd=REPO,
+    )
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    d = _last_json_line(r.stdout)
+    assert d["n_gpus"] == 4 and d["config"]["parallelism"] == "dp4"
+    assert sum(1 for l in r.stdout.splitlines() if l.startswith("{")) == 1
