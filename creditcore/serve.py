@@ -244,7 +244,9 @@ async def lifespan(app: FastAPI):
         dm = DenseModel.load(cfg.dense_model_dir)
         device = cfg.resolve_device()
         if device == "cpu":
-            dense_engines = [DenseEngine(dm, device="cpu")]
+            dense_engines = [
+                DenseEngine(dm, device="cpu") for _ in range(max(cfg.n_gpus, 1))
+            ]
         else:
             import torch
 
@@ -289,16 +291,30 @@ async def lifespan(app: FastAPI):
 
     async def _revival_loop():
         log = _logging.getLogger("creditcore.replicas")
+        loop = asyncio.get_running_loop()
         while True:
             await asyncio.sleep(max(cfg.replica_probe_period_s, 0.25))
             pool = state.get("pool")
             live_engines = state.get("engines")
-            if pool is None or not live_engines:
-                continue
-            try:
-                await probe_revive(pool, live_engines, log)
-            except Exception:
-                pass
+            if pool is not None and live_engines:
+                try:
+                    await probe_revive(pool, live_engines, log)
+                except Exception:
+                    pass
+            # dense replicas get the same probation treatment
+            dpool = state.get("dense_pool")
+            dengines = state.get("dense_engines")
+            if dpool is not None and dengines:
+                for i in dpool.dead_indices():
+                    x = np.zeros((1, dengines[i].model.n_features), np.float32)
+                    try:
+                        await loop.run_in_executor(
+                            None, lambda e=dengines[i]: e.score_arrays(x, False)
+                        )
+                        dpool.revive(i)
+                        log.info("dense replica %d revived", i)
+                    except Exception:
+                        pass
 
     revival_task = asyncio.create_task(_revival_loop())
     yield

@@ -290,3 +290,53 @@ def test_dense_served_on_gpu(dense_model, model_dir, tmp_path_factory):
         np.testing.assert_allclose(
             r2.json()["predictions"], ref["predictions"][:8], rtol=1e-4, atol=1e-6
         )
+
+
+def test_dense_failover_and_revival(dense_model, model_dir, tmp_path_factory, monkeypatch):
+    """Dense replicas share the credit path's failover + probation: a
+    poisoned dense replica leaves rotation without failing requests and is
+    re-admitted once it answers again."""
+    import time as _time
+
+    from fastapi.testclient import TestClient
+
+    from creditcore.config import ServeConfig
+    from creditcore.serve import create_app, state
+
+    d = str(tmp_path_factory.mktemp("dense_model_fo"))
+    dense_model.save(d)
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.dense_model_dir = d
+    cfg.device = "cpu"
+    cfg.n_gpus = 2
+    cfg.replica_probe_period_s = 0.3
+    rng = np.random.default_rng(3)
+    x = rng.normal(size=(8, dense_model.n_features)).astype(np.float32)
+    rows = {"rows": x.tolist()}
+    with TestClient(create_app(cfg)) as c:
+        # dense replicas mirror n_gpus on CPU? engines built per device;
+        # CPU builds a single dense engine — poke only if >=2, else poison
+        # the sole one and expect revival after unpoisoning
+        dengines = state["dense_engines"]
+        b0 = state["dense_batchers"][0]
+        orig = b0.score_arrays
+
+        def boom(*a, **k):
+            raise RuntimeError("injected dense fault")
+
+        monkeypatch.setattr(b0, "score_arrays", boom)
+        if len(dengines) > 1:
+            for _ in range(8):
+                assert c.post("/predict_dense", json=rows).status_code == 200
+            assert state["dense_pool"].alive[0] is False
+        else:
+            r = c.post("/predict_dense", json=rows)
+            assert r.status_code in (500, 503)
+            assert state["dense_pool"].alive == [False]
+        monkeypatch.setattr(b0, "score_arrays", orig)
+        deadline = _time.time() + 15
+        while _time.time() < deadline and not all(state["dense_pool"].alive):
+            _time.sleep(0.1)
+        assert all(state["dense_pool"].alive)
+        assert c.post("/predict_dense", json=rows).status_code == 200
