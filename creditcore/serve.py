@@ -414,24 +414,50 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
                 )
             return rb
 
-        one_minus = (
-            np.float32(1.0) - np.asarray(out["p_vals"], dtype=np.float32)
-        ).astype(np.float64)
-        response = {
-            "predictions": np.asarray(out["predictions"]).tolist(),
-            "outliers": np.asarray(out["outliers"]).tolist(),
-            "feature_drift_batch": dict(zip(FEATURES, one_minus.tolist())),
-        }
+        # merged-flush responses: C serializer (GIL released) beats Python
+        # json.dumps by ~10x on 1024-row responses
+        from .ops import gpu as _gpu
+
+        if _gpu.available():
+            response = bytes(
+                _gpu.ext().build_response_json_arrays(
+                    np.ascontiguousarray(out["predictions"], dtype=np.float64),
+                    np.ascontiguousarray(out["outliers"], dtype=np.float64),
+                    np.ascontiguousarray(out["p_vals"], dtype=np.float64),
+                    FEATURES,
+                )
+            )
+            log_payload = response.decode("utf-8", "replace")
+        else:
+            one_minus = (
+                np.float32(1.0) - np.asarray(out["p_vals"], dtype=np.float32)
+            ).astype(np.float64)
+            response = {
+                "predictions": np.asarray(out["predictions"]).tolist(),
+                "outliers": np.asarray(out["outliers"]).tolist(),
+                "feature_drift_batch": dict(zip(FEATURES, one_minus.tolist())),
+            }
+            log_payload = None
         metrics.observe_request(len(codes), latency_ms)
         if cfg.log_responses:
-            reqlog.log_model_output(
-                cfg.service_name,
-                request_id,
-                response,
-                latency_ms=latency_ms,
-                rows=len(codes),
-                device=f"{engines[idx].device}:{engines[idx].device_index}",
-            )
+            if log_payload is not None:
+                reqlog.log_model_output_raw(
+                    cfg.service_name,
+                    request_id,
+                    log_payload,
+                    latency_ms=latency_ms,
+                    rows=len(codes),
+                    device=f"{engines[idx].device}:{engines[idx].device_index}",
+                )
+            else:
+                reqlog.log_model_output(
+                    cfg.service_name,
+                    request_id,
+                    response,
+                    latency_ms=latency_ms,
+                    rows=len(codes),
+                    device=f"{engines[idx].device}:{engines[idx].device_index}",
+                )
         return response
 
     # The endpoints take the raw body (native JSON fast path) but keep the

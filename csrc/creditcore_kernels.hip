@@ -2196,6 +2196,50 @@ py::bytes build_response_json(py::object pin_outs, int64_t b,
   return py::bytes(out);
 }
 
+// Serialize a /score response from plain arrays (the micro-batcher's
+// merged-flush slices) — same wire bytes as build_response_json but not
+// tied to the session's pinned layout. GIL released during the build.
+py::bytes build_response_json_arrays(py::array_t<double> predictions,
+                                     py::array_t<double> outliers,
+                                     py::array_t<double> pvals,
+                                     py::list feature_names) {
+  const double* proba = predictions.data();
+  const double* outl = outliers.data();
+  const double* pv = pvals.data();
+  const int64_t b = (int64_t)predictions.size();
+  TORCH_CHECK((int64_t)outliers.size() == b, "outliers size mismatch");
+  const int nf = (int)py::len(feature_names);
+  TORCH_CHECK((int)pvals.size() == nf, "pvals size mismatch");
+  std::vector<std::string> names(nf);
+  for (int j = 0; j < nf; ++j) names[j] = py::cast<std::string>(feature_names[j]);
+
+  std::string out;
+  {
+    py::gil_scoped_release nogil;
+    out.reserve((size_t)b * 24 + 2048);
+    out += "{\"predictions\": [";
+    for (int64_t i = 0; i < b; ++i) {
+      if (i) out += ", ";
+      append_double(out, proba[i]);
+    }
+    out += "], \"outliers\": [";
+    for (int64_t i = 0; i < b; ++i) {
+      if (i) out += ", ";
+      out += (outl[i] != 0.0) ? "1.0" : "0.0";
+    }
+    out += "], \"feature_drift_batch\": {";
+    for (int j = 0; j < nf; ++j) {
+      if (j) out += ", ";
+      out += '\"';
+      out += names[j];
+      out += "\": ";
+      append_double(out, (double)(1.0f - (float)pv[j]));
+    }
+    out += "}}";
+  }
+  return py::bytes(out);
+}
+
 // ---------------------------------------------------------------------------
 // The consolidated request path: one call = parse wire JSON -> pinned
 // staging -> graph replay -> drift p-values -> response JSON bytes. No
@@ -2363,6 +2407,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("rs_off"), py::arg("block") = 256, py::arg("ref_lds") = 0);
   m.def("build_response_json", &build_response_json,
         "Serialize the /score response to JSON bytes (C, shortest doubles)");
+  m.def("build_response_json_arrays", &build_response_json_arrays,
+        "Serialize a /score response from plain arrays (merged-flush slices)");
   m.def("drift_pvals_host", &drift_pvals_host,
         "Drift p-values from kernel statistics (chi2 + Pelz-Good K-S), host C");
   py::class_<ScoreSession>(m, "ScoreSession")
