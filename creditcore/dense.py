@@ -189,6 +189,44 @@ def train_dense(
     return model.as_numpy()
 
 
+def parse_dense_body(body: bytes, content_type: str, n_features: int) -> np.ndarray:
+    """Parse a /predict_dense request body. Two wire formats: binary
+    little-endian f32 (8-byte uint32 rows, uint32 cols header — the bulk
+    path) and JSON {"rows": [[...], ...]} (any standard client; nulls
+    become NaN -> median-imputed). Raises ValueError on contract
+    violations (callers map to 422)."""
+    import json
+    import struct
+
+    if content_type.startswith("application/json") or body[:1] in (b"{", b"["):
+        try:
+            doc = json.loads(body)
+        except ValueError:
+            raise ValueError("invalid JSON body")
+        rows = doc.get("rows") if isinstance(doc, dict) else doc
+        if not isinstance(rows, list) or not rows:
+            raise ValueError('expected {"rows": [[...], ...]}')
+        try:
+            x = np.array(
+                [[np.nan if v is None else v for v in r] for r in rows],
+                dtype=np.float32,
+            )
+        except (TypeError, ValueError) as e:
+            raise ValueError(f"bad row: {e}")
+        if x.ndim != 2 or x.shape[1] != n_features:
+            raise ValueError(f"expected {n_features} features per row")
+        return x
+    if len(body) < 8:
+        raise ValueError("missing rows/cols header")
+    rows, cols = struct.unpack("<II", body[:8])
+    if cols != n_features:
+        raise ValueError(f"expected {n_features} features, got {cols}")
+    expect = 8 + rows * cols * 4
+    if rows == 0 or len(body) != expect:
+        raise ValueError("body size mismatch")
+    return np.frombuffer(body, dtype="<f4", offset=8).reshape(rows, cols)
+
+
 class DenseEngine:
     """Scoring engine for the dense family (one replica / GPU).
 

@@ -39,6 +39,9 @@ class RawScoreServer:
         self.batchers = []
         self.drift_sync = None
         self.pool = None
+        self.dense_engines = []
+        self.dense_batchers = []
+        self.dense_pool = None
 
     def _wire_batchers(self, engines) -> list[MicroBatcher]:
         cfg = self.cfg
@@ -91,6 +94,48 @@ class RawScoreServer:
             await b.start()
         self.pool = ReplicaPool(len(self.engines))
 
+        # dense wide-tabular family (BASELINE config 5) on the raw
+        # frontend: same replica/batcher/failover wiring as serve.py
+        self.dense_engines = []
+        self.dense_batchers = []
+        self.dense_pool = None
+        if cfg.dense_model_dir:
+            from .dense import DenseEngine, DenseModel
+
+            dm = DenseModel.load(cfg.dense_model_dir)
+            device = cfg.resolve_device()
+            if device == "cpu":
+                self.dense_engines = [
+                    DenseEngine(dm, device="cpu") for _ in range(max(cfg.n_gpus, 1))
+                ]
+            else:
+                import torch
+
+                n_dev = max(torch.cuda.device_count(), 1)
+                n = cfg.n_gpus or n_dev
+                self.dense_engines = [
+                    DenseEngine(dm, device="cuda", device_index=i % n_dev)
+                    for i in range(n)
+                ]
+
+            def _dscorer(e):
+                def run(x, _nums):
+                    return e.score_arrays(x)
+
+                return run
+
+            self.dense_batchers = [
+                MicroBatcher(
+                    _dscorer(e),
+                    max_rows=cfg.max_batch_rows,
+                    max_wait_us=cfg.batch_wait_us,
+                )
+                for e in self.dense_engines
+            ]
+            for b in self.dense_batchers:
+                await b.start()
+            self.dense_pool = ReplicaPool(len(self.dense_engines))
+
         # probation loop: re-probe dead replicas, re-admit on success
         async def _revival_loop():
             from .serve import probe_revive
@@ -116,6 +161,8 @@ class RawScoreServer:
         self._server.close()
         await self._server.wait_closed()
         for b in self.batchers:
+            await b.close()
+        for b in getattr(self, "dense_batchers", []) or []:
             await b.close()
         if self.cfg.drift_state_path and self.cfg.workers <= 1:
             self.drift_sync.save_state(self.cfg.drift_state_path)
@@ -229,6 +276,51 @@ class RawScoreServer:
                 device=f"{engine.device}:{engine.device_index}",
             )
         return 200, payload
+
+    async def _score_dense(self, body: bytes, ctype: bytes) -> tuple[int, bytes]:
+        """Dense wide-tabular scoring on the raw frontend (binary bulk or
+        JSON rows bodies; micro-batched, replica failover)."""
+        if not self.dense_engines:
+            return 404, b'{"detail": "no dense model configured"}'
+        from .dense import parse_dense_body
+
+        try:
+            x = parse_dense_body(
+                body, ctype.decode("ascii", "replace"),
+                self.dense_engines[0].model.n_features,
+            )
+        except ValueError as e:
+            return 422, json.dumps({"detail": str(e)}).encode()
+        t0 = time.perf_counter()
+        out = None
+        last_exc = None
+        for _ in range(self.dense_pool.attempt_budget):
+            try:
+                idx = self.dense_pool.pick()
+            except RuntimeError:
+                self.metrics.observe_error()
+                return 503, b'{"detail": "no healthy dense replicas"}'
+            try:
+                out = await self.dense_batchers[idx].submit(x, None)
+                self.dense_pool.report_ok(idx)
+                break
+            except Exception as e:
+                self.metrics.observe_error()
+                self.dense_pool.report_fail(idx)
+                last_exc = e
+        if out is None:
+            return 500, json.dumps({"detail": f"scoring failed: {last_exc}"}).encode()
+        self.metrics.observe_request(len(x), (time.perf_counter() - t0) * 1e3)
+        one_minus = (
+            np.float32(1.0) - np.asarray(out["p_vals"], dtype=np.float32)
+        ).astype(np.float64)
+        return 200, json.dumps(
+            {
+                "predictions": np.asarray(out["predictions"]).tolist(),
+                "outliers": np.asarray(out["outliers"]).tolist(),
+                "feature_drift_batch": one_minus.tolist(),
+            }
+        ).encode()
 
     async def _reload(self, body: bytes) -> tuple[int, bytes]:
         """Hot model swap (per worker): build new engines first, then swap
@@ -347,6 +439,7 @@ class RawScoreServer:
                 keep_alive = True
                 chunked = False
                 admin_hdrs = {}
+                ctype_in = b""
                 for h in lines[1:]:
                     if not h:
                         continue
@@ -367,6 +460,8 @@ class RawScoreServer:
                         keep_alive = False
                     elif lk in (b"authorization", b"x-admin-token"):
                         admin_hdrs[lk] = v.strip()
+                    elif lk == b"content-type":
+                        ctype_in = v.strip()
                     elif lk == b"transfer-encoding":
                         te = v.strip().lower()
                         if te == b"chunked":
@@ -393,6 +488,8 @@ class RawScoreServer:
                 ppath = path.partition(b"?")[0]
                 if method == b"POST" and ppath in (b"/score", b"/predict"):
                     status, payload = await self._score(body)
+                elif method == b"POST" and ppath == b"/predict_dense":
+                    status, payload = await self._score_dense(body, ctype_in)
                 elif method == b"POST" and ppath == b"/admin/reload":
                     from .serve import admin_authorized
 

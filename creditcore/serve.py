@@ -538,48 +538,12 @@ def create_app(cfg: ServeConfig | None = None) -> FastAPI:
         return state["metrics"].snapshot()
 
     def _parse_dense_body(body: bytes, content_type: str, n_features: int):
-        """Two wire formats: binary little-endian f32 (8-byte uint32 rows,
-        uint32 cols header — the bulk path) and JSON {"rows": [[...], ...]}
-        (any standard client; nulls become NaN -> median-imputed)."""
-        import struct
+        from .dense import parse_dense_body
 
-        if content_type.startswith("application/json") or (
-            body[:1] in (b"{", b"[")
-        ):
-            try:
-                doc = json.loads(body)
-            except ValueError:
-                raise HTTPException(status_code=422, detail="invalid JSON body")
-            rows = doc.get("rows") if isinstance(doc, dict) else doc
-            if not isinstance(rows, list) or not rows:
-                raise HTTPException(
-                    status_code=422, detail='expected {"rows": [[...], ...]}'
-                )
-            try:
-                x = np.array(
-                    [[np.nan if v is None else v for v in r] for r in rows],
-                    dtype=np.float32,
-                )
-            except (TypeError, ValueError) as e:
-                raise HTTPException(status_code=422, detail=f"bad row: {e}")
-            if x.ndim != 2 or x.shape[1] != n_features:
-                raise HTTPException(
-                    status_code=422,
-                    detail=f"expected {n_features} features per row",
-                )
-            return x
-        if len(body) < 8:
-            raise HTTPException(status_code=422, detail="missing rows/cols header")
-        rows, cols = struct.unpack("<II", body[:8])
-        if cols != n_features:
-            raise HTTPException(
-                status_code=422,
-                detail=f"expected {n_features} features, got {cols}",
-            )
-        expect = 8 + rows * cols * 4
-        if rows == 0 or len(body) != expect:
-            raise HTTPException(status_code=422, detail="body size mismatch")
-        return np.frombuffer(body, dtype="<f4", offset=8).reshape(rows, cols)
+        try:
+            return parse_dense_body(body, content_type, n_features)
+        except ValueError as e:
+            raise HTTPException(status_code=422, detail=str(e))
 
     @app.post("/predict_dense")
     async def predict_dense(request: Request):

@@ -378,3 +378,74 @@ def test_chunked_extensions_trailers_and_limits(raw_url):
         b"POST /score HTTP/1.1\r\nHost: x\r\nTransfer-Encoding: gzip\r\n\r\n"
     )
     assert send(gz).startswith(b"HTTP/1.1 501")
+
+
+def test_dense_on_raw_frontend(model_dir, tmp_path_factory):
+    """/predict_dense on the raw frontend: binary and JSON bodies through
+    the micro-batcher, parity with the engine."""
+    import socket
+    import struct
+    import threading
+
+    import httpx
+    import numpy as np
+
+    from creditcore.config import ServeConfig
+    from creditcore.dense import DenseEngine, train_dense
+    from creditcore.rawserve import RawScoreServer
+
+    model = train_dense(
+        n_rows=50_000, n_feats=32, ref_rows=5_000, epochs=1,
+        batch_rows=8192, device="cpu", seed=5, log=lambda *a: None,
+    )
+    d = str(tmp_path_factory.mktemp("dense_raw"))
+    model.save(d)
+
+    s = socket.socket(); s.bind(("127.0.0.1", 0)); port = s.getsockname()[1]; s.close()
+    cfg = ServeConfig()
+    cfg.model_directory = model_dir
+    cfg.dense_model_dir = d
+    cfg.device = "cpu"
+    cfg.host = "127.0.0.1"
+    cfg.port = port
+
+    loop = asyncio.new_event_loop()
+    server = RawScoreServer(cfg)
+    started = threading.Event()
+
+    def run():
+        asyncio.set_event_loop(loop)
+
+        async def go():
+            await server.start()
+            started.set()
+            await server._server.serve_forever()
+
+        try:
+            loop.run_until_complete(go())
+        except (asyncio.CancelledError, RuntimeError):
+            pass
+
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    assert started.wait(timeout=120)
+    try:
+        url = f"http://127.0.0.1:{port}"
+        rng = np.random.default_rng(2)
+        x = rng.normal(size=(16, 32)).astype(np.float32)
+        body = struct.pack("<II", *x.shape) + x.tobytes()
+        r = httpx.post(f"{url}/predict_dense", content=body,
+                       headers={"Content-Type": "application/octet-stream"},
+                       timeout=60)
+        assert r.status_code == 200, r.text
+        ref = DenseEngine(model, device="cpu").score_arrays(x)
+        np.testing.assert_allclose(r.json()["predictions"], ref["predictions"],
+                                   rtol=1e-6)
+        r2 = httpx.post(f"{url}/predict_dense", json={"rows": x[:3].tolist()},
+                        timeout=60)
+        assert r2.status_code == 200
+        assert httpx.post(f"{url}/predict_dense", content=b"xx",
+                          timeout=30).status_code == 422
+    finally:
+        loop.call_soon_threadsafe(loop.stop)
+        t.join(timeout=30)
