@@ -354,9 +354,12 @@ def main():
     if device == "cuda":
         t_burn = time.perf_counter()
         burn = 0
-        # long enough to reach thermal/clock steady state (~0.5 s of
-        # continuous graph replays), still bounded for big models
-        while burn < 4096 and time.perf_counter() - t_burn < 3.0:
+        # long enough to reach thermal/clock steady state: boxes measure
+        # the FIRST invocation ~15-25% slower than the second even after
+        # 0.5 s of replays, so burn a full ~3 s of continuous graph
+        # replays (still bounded by step count for big models)
+        burn_s = float(os.environ.get("CREDITCORE_BENCH_BURN_S", "3.0"))
+        while burn < 32768 and time.perf_counter() - t_burn < burn_s:
             run_steps(64)
             burn += 64
         print(f"[bench] burn-in: {burn} steps in "
