@@ -202,11 +202,10 @@ def main():
             )
             if r.returncode != 0:
                 raise RuntimeError("packed-model build subprocess failed")
-            # post-training cooldown: ~60 s of all-core sklearn training
-            # leaves the package hot and the cores off their boost clocks;
-            # runs that trained immediately before the timed loop measured
-            # 15-40% slower than runs starting from a cached model
-            cooldown = float(os.environ.get("CREDITCORE_BENCH_COOLDOWN_S", "10"))
+            # optional post-training cooldown (A/B'd: no measurable effect
+            # — the "first invocation slower" pattern was run-to-run box
+            # variance, ±15-20% on 20-step windows; kernel_tuning.md)
+            cooldown = float(os.environ.get("CREDITCORE_BENCH_COOLDOWN_S", "0"))
             if cooldown > 0:
                 print(f"[bench] post-training cooldown {cooldown:.0f}s",
                       file=sys.stderr)
