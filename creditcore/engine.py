@@ -111,9 +111,8 @@ class ScoringEngine:
         # Drift is a batch-population statistic; cap its sample at the K-S
         # kernel's LDS sort capacity (predictions still cover every row).
         drift_now = with_drift and b <= self.DRIFT_MAX_ROWS
-        g["np_codes"][0][:b] = codes
-        g["np_nums"][0][:b] = nums
-        g["sess"].score(b, drift_now, True)  # blocks; GIL released
+        # one C call: pinned staging memcpy (GIL released) + graph replay
+        g["sess"].submit_arrays(codes, nums, 0, drift_now, True)
         flat = g["np_outs"][0].reshape(-1)  # b-packed: proba | iscore | outlier
         out = {
             "predictions": flat[:b].copy(),
@@ -271,12 +270,12 @@ class ScoringEngine:
             }
         g = self._gpu
         self._ensure_capacity(b)
-        g["np_codes"][0][:b] = codes
-        g["np_nums"][0][:b] = nums
         nb = b
         if b <= self.DRIFT_MAX_ROWS:
-            g["sess"].score(b, True, True)
+            g["sess"].submit_arrays(codes, nums, 0, True, True)
         else:
+            g["np_codes"][0][:b] = codes
+            g["np_nums"][0][:b] = nums
             # Oversized batch: drift is a batch-population statistic, so run
             # the capped-sample drift pass FIRST (only with_drift passes
             # write the pinned drift blob), then the full batch without
@@ -307,10 +306,7 @@ class ScoringEngine:
         b = len(codes)
         self._ensure_capacity(b)
         assert b <= self.DRIFT_MAX_ROWS, "pipelined path caps at DRIFT_MAX_ROWS"
-        s = slot & 1
-        g["np_codes"][s][:b] = codes
-        g["np_nums"][s][:b] = nums
-        g["sess"].score(b, True, False, s)
+        g["sess"].submit_arrays(codes, nums, slot & 1, True, False)
         return b
 
     def finish_slot(self, slot: int, b: int) -> dict:
