@@ -322,12 +322,15 @@ def main():
                 )
                 stream["i"] += 1
 
+        phases = [] if os.environ.get("CREDITCORE_BENCH_PHASES") else None
+
         def run_steps(k: int, step_times=None):
             outs = []
             _prime()
             t_prev = time.perf_counter()
             pending = None  # (slot, rows, nums) awaiting epilogue
             for i in range(k):
+                t_a = time.perf_counter() if phases is not None else 0.0
                 codes, nums = q.popleft().result()
                 _prime()  # replacement encode overlaps this step's GPU work
                 if use_slots:
@@ -335,9 +338,18 @@ def main():
                     # epilogue (p-values + response serialization) while the
                     # GPU executes step i — double-buffered pinned slots
                     slot = i & 1
+                    t_b = time.perf_counter() if phases is not None else 0.0
                     b = engine.submit_encoded_slot(codes, nums, slot)
+                    t_c = time.perf_counter() if phases is not None else 0.0
                     if pending is not None:
                         out = engine.finish_slot(pending[0], pending[1])
+                        if phases is not None and step_times is not None:
+                            t_d = time.perf_counter()
+                            phases.append(
+                                (round((t_b - t_a) * 1e6, 1),   # encode wait+prime
+                                 round((t_c - t_b) * 1e6, 1),   # submit (stage+launch)
+                                 round((t_d - t_c) * 1e6, 1))   # finish (wait+epilogue)
+                            )
                         t_prev = _post_step(out, pending[2], len(outs), outs, step_times, t_prev)
                     pending = (slot, b, nums)
                 elif with_drift:
@@ -466,6 +478,13 @@ def main():
     if rank == 0 and args.dump_steps:
         with open(args.dump_steps, "w") as f:
             json.dump([round(t * 1e6, 2) for t in step_times], f)
+    ph_path = os.environ.get("CREDITCORE_BENCH_PHASES")
+    if rank == 0 and ph_path:
+        try:
+            with open(ph_path, "w") as f:
+                json.dump(phases, f)
+        except NameError:
+            pass  # latency mode has no phases
 
     if distributed:
         import torch.distributed as dist
