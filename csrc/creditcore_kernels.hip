@@ -1180,7 +1180,13 @@ inline void sync_stream(hipStream_t s) {
 }
 
 inline void sync_event(hipEvent_t ev) {
-  if (spin_sync_enabled()) {
+  // Event waits run on the EPILOGUE thread concurrently with the main
+  // thread's hipGraphLaunch: a hipEventQuery spin there hammers the HIP
+  // runtime's internal locks and was measured to double the launch cost
+  // (submit 51 -> 96 us at b=1024). Block by default; CREDITCORE_SPIN_EVENT=1
+  // restores the spin for single-threaded callers.
+  static const bool spin = (std::getenv("CREDITCORE_SPIN_EVENT") != nullptr);
+  if (spin) {
     hipError_t e;
     while ((e = hipEventQuery(ev)) == hipErrorNotReady) {}
     if (e != hipSuccess) HIP_CHECK(e);
