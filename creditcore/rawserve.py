@@ -363,13 +363,37 @@ class RawScoreServer:
             return 200, self.metrics.prometheus().encode(), b"text/plain; version=0.0.4"
         if path == b"/healthz":
             any_alive = any(self.pool.alive)
-            return 200, json.dumps(
-                {
-                    "status": "ok" if any_alive else "dead",
-                    "engines": len(self.engines),
-                    "alive": self.pool.alive,
-                }
-            ).encode()
+            body = {
+                "status": "ok" if any_alive else "dead",
+                "engines": len(self.engines),
+                "alive": self.pool.alive,
+            }
+            if b"deep" in query and any_alive:
+                # active probe parity with the FastAPI frontend: score the
+                # schema-default record on every live replica
+                from .pack import encode_batch
+                from .schema import LoanApplicant
+
+                codes, nums = encode_batch(
+                    [LoanApplicant().__dict__], self.engines[0].packed.vocabs
+                )
+                loop = asyncio.get_running_loop()
+                probes = []
+                for i, e in enumerate(self.engines):
+                    if not self.pool.alive[i]:
+                        probes.append("dead")
+                        continue
+                    try:
+                        await loop.run_in_executor(
+                            None, lambda e=e: e.score_arrays(codes, nums, False)
+                        )
+                        probes.append("ok")
+                    except Exception as exc:
+                        self.pool.report_fail(i)
+                        probes.append(f"failed: {exc}")
+                        body["status"] = "degraded"
+                body["probe"] = probes
+            return 200, json.dumps(body).encode()
         if path == b"/metrics":
             return 200, json.dumps(self.metrics.snapshot()).encode()
         if path == b"/drift":

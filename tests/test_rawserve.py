@@ -449,3 +449,58 @@ def test_dense_on_raw_frontend(model_dir, tmp_path_factory):
     finally:
         loop.call_soon_threadsafe(loop.stop)
         t.join(timeout=30)
+
+
+def test_deep_health_probe_on_raw_frontend(raw_url):
+    import httpx
+
+    body = httpx.get(f"{raw_url}/healthz?deep=1", timeout=60).json()
+    assert body["status"] == "ok"
+    assert body["probe"] == ["ok"]
+
+
+def test_chunked_random_splits_property(raw_url):
+    """Property-style: any chunk segmentation of a valid body must decode
+    to the same response as the Content-Length path."""
+    import socket
+
+    import httpx
+
+    from creditcore.data import make_request_batch
+
+    body = json.dumps(make_request_batch(8, seed=9)).encode()
+    want = httpx.post(f"{raw_url}/score", content=body,
+                      headers={"Content-Type": "application/json"},
+                      timeout=60).json()
+    host, port = raw_url.rsplit("/", 1)[-1].split(":")
+    import random
+
+    rng = random.Random(1)
+    for trial in range(5):
+        cuts = sorted(rng.sample(range(1, len(body)), rng.randint(1, 12)))
+        parts, lo = [], 0
+        for c in cuts + [len(body)]:
+            parts.append(body[lo:c])
+            lo = c
+        msg = (b"POST /score HTTP/1.1\r\nHost: x\r\n"
+               b"Content-Type: application/json\r\n"
+               b"Transfer-Encoding: chunked\r\n\r\n")
+        for p in parts:
+            msg += format(len(p), "x").encode() + b"\r\n" + p + b"\r\n"
+        msg += b"0\r\n\r\n"
+        with socket.create_connection((host, int(port)), timeout=30) as s:
+            s.sendall(msg)
+            s.settimeout(30)
+            out = b""
+            while b"\r\n\r\n" not in out:
+                out += s.recv(65536)
+            head, _, rest = out.partition(b"\r\n\r\n")
+            clen = 0
+            for line in head.split(b"\r\n"):
+                if line.lower().startswith(b"content-length:"):
+                    clen = int(line[15:])
+            while len(rest) < clen:
+                rest += s.recv(65536)
+        assert head.startswith(b"HTTP/1.1 200"), head[:60]
+        got = json.loads(rest[:clen])
+        assert got["predictions"] == want["predictions"], trial
