@@ -287,3 +287,19 @@ def test_native_code_is_loaded(gpu_engine):
 
     assert "creditcore" in ccore.__file__
     assert ccore.__file__.endswith(".so")
+
+
+def test_submit_arrays_strided_inputs(gpu_engine, packed):
+    """submit_arrays must handle non-contiguous inputs (row-wise copy
+    path) identically to contiguous ones."""
+    b = 256
+    codes, nums = _random_encoded(packed, 2 * b, seed=21)
+    # every other row: stride-2 views, non-contiguous
+    cs, ns = codes[::2], nums[::2]
+    assert not cs.flags["C_CONTIGUOUS"]
+    out_strided = gpu_engine.score_arrays(cs, ns, with_drift=False)
+    out_contig = gpu_engine.score_arrays(
+        np.ascontiguousarray(cs), np.ascontiguousarray(ns), with_drift=False
+    )
+    np.testing.assert_array_equal(out_strided["predictions"], out_contig["predictions"])
+    np.testing.assert_array_equal(out_strided["outliers"], out_contig["outliers"])
