@@ -10,8 +10,10 @@ validation); enable this one with `creditcore serve --raw-http` where
 throughput matters.
 
 Endpoints: POST /score, /predict (wire-format fast path with pydantic
-fallback semantics preserved via the same `_encode` logic), GET /healthz,
-/metrics, /drift.
+fallback semantics preserved via the same `_encode` logic),
+POST /predict_dense (binary bulk or JSON rows), POST /admin/reload
+(loopback/token gated), GET /healthz[?deep], /metrics, /drift. Bodies may
+use Content-Length or chunked transfer encoding.
 """
 
 from __future__ import annotations
