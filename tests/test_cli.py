@@ -110,3 +110,35 @@ def test_lint_gate_clean():
         capture_output=True, text=True, timeout=120,
     )
     assert r.returncode == 0, r.stdout + r.stderr
+
+
+def test_config_new_round2_fields(monkeypatch):
+    """Round-2 config fields parse from env and CLI with the same
+    precedence as the rest of ServeConfig."""
+    from creditcore.config import ServeConfig
+
+    monkeypatch.setenv("CREDITCORE_ADMIN_TOKEN", "envtok")
+    monkeypatch.setenv("CREDITCORE_REPLICA_PROBE_PERIOD_S", "2.5")
+    monkeypatch.setenv("CREDITCORE_DRIFT_MAX_BATCH", "4096")
+    cfg = ServeConfig()
+    assert cfg.admin_token == "envtok"
+    assert cfg.replica_probe_period_s == 2.5
+    assert cfg.drift_max_batch == 4096
+    # CLI overrides env
+    cfg2 = ServeConfig.from_args(
+        ["--admin-token", "clitok", "--replica-probe-period-s", "7"]
+    )
+    assert cfg2.admin_token == "clitok"
+    assert cfg2.replica_probe_period_s == 7.0
+    assert cfg2.drift_max_batch == 4096  # env still applies where no flag
+
+
+def test_engine_drift_cap_clamps_to_hardware(packed):
+    """drift_max_batch above the 16384 LDS ceiling clamps instead of
+    launching an unsupported K-S shape."""
+    from creditcore.engine import ScoringEngine
+
+    e = ScoringEngine(packed, device="cpu", drift_max_rows=99999)
+    assert e.DRIFT_MAX_ROWS == ScoringEngine.HW_DRIFT_MAX_ROWS == 16384
+    e2 = ScoringEngine(packed, device="cpu", drift_max_rows=2048)
+    assert e2.DRIFT_MAX_ROWS == 2048
