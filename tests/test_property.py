@@ -144,3 +144,67 @@ def test_driftsync_torch_and_numpy_paths_agree(packed, seed, rows, nan_frac):
     b.accumulate(cat_hist, nums)  # numpy fast path
     np.testing.assert_array_equal(a.local.numpy(), b.local.numpy())
     assert a.batches == b.batches == 1
+
+
+@settings(max_examples=80, deadline=None)
+@given(
+    body=st.binary(max_size=256),
+    ctype=st.sampled_from(["application/json", "application/octet-stream", ""]),
+)
+def test_parse_dense_body_never_crashes(body, ctype):
+    """Fuzz: parse_dense_body on arbitrary bytes either returns a valid
+    [B, F] float32 array or raises ValueError — never any other exception."""
+    from creditcore.dense import parse_dense_body
+
+    try:
+        x = parse_dense_body(body, ctype, 8)
+    except ValueError:
+        return
+    assert x.ndim == 2 and x.shape[1] == 8 and x.dtype == np.float32
+
+
+@settings(max_examples=40, deadline=None)
+@given(
+    rows=st.integers(min_value=1, max_value=9),
+    cols=st.integers(min_value=1, max_value=12),
+    seed=st.integers(min_value=0, max_value=2**16),
+)
+def test_parse_dense_body_binary_roundtrip(rows, cols, seed):
+    import struct
+
+    from creditcore.dense import parse_dense_body
+
+    rng = np.random.default_rng(seed)
+    x = rng.normal(size=(rows, cols)).astype("<f4")
+    body = struct.pack("<II", rows, cols) + x.tobytes()
+    got = parse_dense_body(body, "application/octet-stream", cols)
+    np.testing.assert_array_equal(got, x)
+    # wrong feature count must be rejected
+    with pytest.raises(ValueError):
+        parse_dense_body(body, "application/octet-stream", cols + 1)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    data=st.lists(
+        st.lists(
+            st.one_of(st.none(), st.floats(allow_nan=False, allow_infinity=False,
+                                           width=32)),
+            min_size=5, max_size=5,
+        ),
+        min_size=1, max_size=6,
+    )
+)
+def test_parse_dense_body_json_nulls_to_nan(data):
+    import json as _json
+
+    from creditcore.dense import parse_dense_body
+
+    x = parse_dense_body(_json.dumps({"rows": data}).encode(),
+                         "application/json", 5)
+    for i, row in enumerate(data):
+        for j, v in enumerate(row):
+            if v is None:
+                assert np.isnan(x[i, j])
+            else:
+                assert x[i, j] == np.float32(v)
