@@ -260,3 +260,29 @@ def test_cpu_engine_json_body_fills_schema_defaults(packed, monkeypatch):
         eng.encode_json_body(b'{"not": "a list"}')
     with _pytest.raises(ValueError):
         eng.encode_json_body(b'[1, 2]')
+
+
+def test_c_serializer_arrays_roundtrip(ext):
+    """build_response_json_arrays (merged-flush path) must reproduce every
+    double bit-exactly and match the reference (1 - p) float32 semantics."""
+    import json as _json
+
+    from creditcore.schema import FEATURES
+
+    vals = [0.0, 1.0, 2 / 3, 1e-300, 5e-324, 0.49999999999999994, 1e16]
+    preds = np.array(vals, dtype=np.float64)
+    outl = np.array([i % 2 for i in range(len(vals))], dtype=np.float64)
+    pvals = np.linspace(0.0, 1.0, 23)
+    raw = ext.build_response_json_arrays(preds, outl, pvals, FEATURES)
+    doc = _json.loads(raw)
+    assert doc["predictions"] == vals
+    assert doc["outliers"] == [float(i % 2) for i in range(len(vals))]
+    for f, p in zip(FEATURES, pvals):
+        assert doc["feature_drift_batch"][f] == float(np.float32(1.0) - np.float32(p))
+    # size mismatches are rejected loudly
+    import pytest as _pytest
+
+    with _pytest.raises(Exception):
+        ext.build_response_json_arrays(preds, outl[:-1], pvals, FEATURES)
+    with _pytest.raises(Exception):
+        ext.build_response_json_arrays(preds, outl, pvals[:-1], FEATURES)
