@@ -84,6 +84,8 @@ def broadcast_packed(
             name: list(getattr(packed, name).shape) for name, _ in _ARRAY_FIELDS
         }
         meta["has_lin"] = packed.lin_weight is not None
+        if meta["has_lin"]:
+            meta["lin_shape"] = list(packed.lin_weight.shape)
         obj = [meta]
     else:
         obj = [None]
@@ -102,13 +104,25 @@ def broadcast_packed(
         dist.broadcast(t, src=src, group=group)
         tensors[name] = t
 
+    # optional linear-predict weights (dense/linear family models)
+    lin_weight = None
+    if meta["has_lin"]:
+        if is_src:
+            lw = torch.from_numpy(
+                np.ascontiguousarray(packed.lin_weight.astype(np.float32, copy=False))
+            ).to(device)
+        else:
+            lw = torch.empty(meta["lin_shape"], dtype=torch.float32, device=device)
+        dist.broadcast(lw, src=src, group=group)
+        lin_weight = lw.cpu().numpy()
+
     if is_src:
         return packed
     kw = {name: tensors[name].cpu().numpy() for name, _ in _ARRAY_FIELDS}
     for f in _META_FIELDS:
         kw[f] = meta[f]
     kw.pop("meta", None)
-    return PackedModel(lin_weight=None, meta=meta["meta"], **kw)
+    return PackedModel(lin_weight=lin_weight, meta=meta["meta"], **kw)
 
 
 class DriftSync:
@@ -162,8 +176,14 @@ class DriftSync:
         import torch
 
         if not torch.is_tensor(nums):
-            self._accumulate_np(np.asarray(cat_hist), np.asarray(nums))
-            return
+            if self.local.device.type == "cpu":
+                self._accumulate_np(np.asarray(cat_hist), np.asarray(nums))
+                return
+            # device-resident accumulator: route numpy inputs through the
+            # torch path (the numpy fast path writes through .numpy(),
+            # which only exists for CPU tensors)
+            cat_hist = torch.from_numpy(np.ascontiguousarray(cat_hist))
+            nums = torch.from_numpy(np.ascontiguousarray(nums))
         self.local[: self.C] += cat_hist.to(self.local.device, torch.int64)
         x = nums.to(self.edges.device).t().contiguous()  # [N_NUM, B]
         x = torch.where(torch.isnan(x), self.medians_t[:, None], x)

@@ -192,3 +192,32 @@ def test_driftsync_cross_process_publish(packed, tmp_path):
     a.merge_published(d)
     snap = a.snapshot()
     assert snap["rows"] == 120  # both workers' numeric rows visible
+
+
+def _lin_broadcast_worker(rank: int, world: int, port: int, npz_path: str, q):
+    try:
+        import torch.distributed as dist
+
+        from creditcore.pack import PackedModel
+        from creditcore.parallel import broadcast_packed
+
+        _init(rank, world, port)
+        if rank == 0:
+            packed = PackedModel.load(npz_path)
+            packed.lin_weight = np.arange(40, dtype=np.float32)
+            packed.lin_bias = 0.25
+        else:
+            packed = None
+        got = broadcast_packed(packed, device="cpu", src=0)
+        q.put((rank, got.lin_weight.tolist(), float(got.lin_bias)))
+        dist.destroy_process_group()
+    except Exception as e:
+        q.put((rank, f"ERROR: {type(e).__name__}: {e}", None))
+
+
+def test_broadcast_packed_carries_linear_weights(packed_npz):
+    """lin_weight must travel with the broadcast (it was silently dropped
+    for non-source ranks before round 2's fix)."""
+    res = _run_workers(_lin_broadcast_worker, packed_npz, 29517)
+    assert res[0][0] == res[1][0] == list(map(float, range(40)))
+    assert res[0][1] == res[1][1] == 0.25
