@@ -221,3 +221,27 @@ def test_broadcast_packed_carries_linear_weights(packed_npz):
     res = _run_workers(_lin_broadcast_worker, packed_npz, 29517)
     assert res[0][0] == res[1][0] == list(map(float, range(40)))
     assert res[0][1] == res[1][1] == 0.25
+
+
+def test_driftsync_load_state_rejects_corruption(packed, tmp_path):
+    """Corrupted or layout-mismatched drift-state files start fresh
+    (return False) instead of crashing the service at boot."""
+    import torch
+
+    from creditcore.parallel import DriftSync
+
+    ds = DriftSync(packed, device="cpu", n_bins=16)
+    p = str(tmp_path / "drift.npz")
+    # corrupt bytes
+    open(p, "wb").write(b"not an npz at all")
+    assert ds.load_state(p) is False
+    # wrong bin layout
+    ds2 = DriftSync(packed, device="cpu", n_bins=8)
+    ds2.save_state(p)
+    assert ds.load_state(p) is False
+    # good round trip
+    ds.local += 3
+    ds.save_state(p)
+    ds3 = DriftSync(packed, device="cpu", n_bins=16)
+    assert ds3.load_state(p) is True
+    assert torch.equal(ds3.local, ds.local)
